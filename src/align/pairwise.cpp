@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 #include <vector>
@@ -204,37 +205,66 @@ std::string align_global_cigar(const char* q, uint32_t q_len, const char* t, uin
   ColumnStore cs;
   myers_nw(q, q_len, t, t_len, &cs);
 
-  // Backward walk from (q_len-1, t_len-1), edlib move priority: up ('I',
+  // Backward walk from (q_len-1, t_len-1) reproducing edlib's move priority.
+  // Priority order is configurable for golden-parity tuning via
+  // RGA_TRACEBACK_ORDER (a permutation of "IDM"); default "IDM" = up ('I',
   // consume query), then left ('D', consume target), then diagonal ('M').
+  static const char* order_env = getenv("RGA_TRACEBACK_ORDER");
+  const char* order = order_env != nullptr ? order_env : "IDM";
+
   std::string ops;  // reversed op chars
   ops.reserve(q_len + t_len);
   int64_t i = q_len - 1;
   int64_t c = t_len - 1;
   int32_t v = cell_value(cs, c, i);
   while (i >= 0 || c >= 0) {
-    if (i >= 0) {
-      int32_t up = cell_value(cs, c, i - 1);
-      if (up + 1 == v) {
-        ops.push_back('I');
-        --i;
-        v = up;
-        continue;
+    bool moved = false;
+    for (const char* o = order; *o != '\0' && !moved; ++o) {
+      switch (*o) {
+        case 'I':
+          if (i >= 0) {
+            int32_t up = cell_value(cs, c, i - 1);
+            if (up + 1 == v) {
+              ops.push_back('I');
+              --i;
+              v = up;
+              moved = true;
+            }
+          }
+          break;
+        case 'D':
+          if (c >= 0) {
+            int32_t left = cell_value(cs, c - 1, i);
+            if (left + 1 == v) {
+              ops.push_back('D');
+              --c;
+              v = left;
+              moved = true;
+            }
+          }
+          break;
+        case 'M':
+          if (i >= 0 && c >= 0) {
+            int32_t diag = cell_value(cs, c - 1, i - 1);
+            int32_t step = (q[i] == t[c]) ? 0 : 1;
+            if (diag + step == v) {
+              ops.push_back('M');  // match or mismatch; standard CIGAR merges both
+              --i;
+              --c;
+              v = diag;
+              moved = true;
+            }
+          }
+          break;
       }
     }
-    if (c >= 0) {
-      int32_t left = cell_value(cs, c - 1, i);
-      if (left + 1 == v) {
-        ops.push_back('D');
-        --c;
-        v = left;
-        continue;
-      }
+    if (!moved) {
+      // Fallback diagonal (cannot happen for a consistent DP).
+      ops.push_back('M');
+      --i;
+      --c;
+      v = cell_value(cs, c, i);
     }
-    int32_t diag = cell_value(cs, c - 1, i - 1);
-    ops.push_back('M');  // match or mismatch; standard CIGAR merges both
-    --i;
-    --c;
-    v = diag;
   }
 
   // Collapse the reversed op string into CIGAR runs (forward order).

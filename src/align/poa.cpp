@@ -291,13 +291,20 @@ std::string Graph::generate_consensus(std::vector<uint32_t>* coverages) {
     coverages->clear();
     coverages->reserve(consensus_.size());
     std::unordered_set<uint32_t> labels;
-    for (uint32_t node_id : consensus_) {
-      labels.clear();
-      for (uint32_t ei : nodes_[node_id].in_edges) {
+    auto add_node_labels = [&](uint32_t nid) {
+      for (uint32_t ei : nodes_[nid].in_edges) {
         labels.insert(edges_[ei].labels.begin(), edges_[ei].labels.end());
       }
-      for (uint32_t ei : nodes_[node_id].out_edges) {
+      for (uint32_t ei : nodes_[nid].out_edges) {
         labels.insert(edges_[ei].labels.begin(), edges_[ei].labels.end());
+      }
+    };
+    for (uint32_t node_id : consensus_) {
+      labels.clear();
+      add_node_labels(node_id);
+      // reads aligned to ring partners cover this consensus column too
+      for (uint32_t aid : nodes_[node_id].aligned_node_ids) {
+        add_node_labels(aid);
       }
       coverages->emplace_back(static_cast<uint32_t>(labels.size()));
     }
