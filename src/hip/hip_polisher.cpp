@@ -1,0 +1,150 @@
+// MI355X pipeline orchestration: devices, batch workers, pull-model window
+// queue, per-item CPU fallback.
+// Capability parity: reference src/cuda/cudapolisher.cpp — device
+// enumeration + warm-up, 90%-free-memory batch sizing, one host thread per
+// batch pulling from a global mutex-guarded queue, failed windows re-polished
+// on CPU, serial ordered merge identical to the CPU path.
+#include <hip/hip_runtime.h>
+
+#include <atomic>
+#include <memory>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "core/polisher.hpp"
+#include "hip/hip_common.hpp"
+#include "hip/poa_batch.hpp"
+
+namespace rga {
+
+class HipPolisher : public Polisher {
+ public:
+  HipPolisher(std::unique_ptr<SequenceParser> sparser, std::unique_ptr<OverlapParser> oparser,
+              std::unique_ptr<SequenceParser> tparser, PolisherConfig config)
+      : Polisher(std::move(sparser), std::move(oparser), std::move(tparser), config) {
+    int n = hip::device_count();
+    if (n < 1) {
+      fprintf(stderr, "[racon::HipPolisher] error: no HIP devices available!\n");
+      exit(1);
+    }
+    for (int d = 0; d < n; ++d) {
+      RGA_HIP_CHECK(hipSetDevice(d));
+      RGA_HIP_CHECK(hipFree(nullptr));  // create the context up front
+      devices_.emplace_back(d);
+    }
+    fprintf(stderr, "[racon::HipPolisher] using %d GPU(s)\n", n);
+  }
+
+  void polish(std::vector<std::unique_ptr<Sequence>>& dst, bool drop_unpolished) override {
+    if (config_.poa_batches < 1) {
+      Polisher::polish(dst, drop_unpolished);
+      return;
+    }
+
+    logger_->log();
+
+    // one batch object per (device, batch) pair, reference cudapolisher.cpp:228-240
+    std::vector<std::unique_ptr<hip::PoaBatch>> batches;
+    for (int d : devices_) {
+      RGA_HIP_CHECK(hipSetDevice(d));
+      size_t free_mem = 0, total_mem = 0;
+      RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
+      size_t budget = free_mem * 9 / 10 / config_.poa_batches;
+      for (uint32_t b = 0; b < config_.poa_batches; ++b) {
+        batches.emplace_back(std::make_unique<hip::PoaBatch>(
+            d, budget, config_.match, config_.mismatch, config_.gap, config_.banded_poa,
+            200 /* MAX_DEPTH_PER_WINDOW, reference cudapolisher.cpp:226 */));
+      }
+    }
+
+    std::vector<bool> polished(windows_.size(), false);
+    std::vector<bool> gpu_handled(windows_.size(), false);
+    std::mutex queue_mutex;
+    uint64_t next_window = 0;
+    std::mutex status_mutex;
+
+    auto worker = [&](hip::PoaBatch* batch) {
+      std::vector<uint64_t> batch_indices;
+      while (true) {
+        batch_indices.clear();
+        batch->reset();
+        {
+          std::lock_guard<std::mutex> lock(queue_mutex);
+          while (next_window < windows_.size()) {
+            auto& w = windows_[next_window];
+            if (w->num_layers() < 3) {
+              // backbone copy, never worth a GPU trip (reference semantics:
+              // consensus = backbone, unpolished)
+              w->set_consensus(std::string(w->sequence(0).first, w->sequence(0).second));
+              ++next_window;
+              continue;
+            }
+            bool never_fits = false;
+            if (batch->add_window(w, &never_fits)) {
+              batch_indices.emplace_back(next_window);
+              ++next_window;
+            } else if (never_fits) {
+              ++next_window;  // leave for the CPU fallback pass
+            } else {
+              break;  // batch full; leave remaining windows for the next round
+            }
+          }
+        }
+        if (batch_indices.empty()) {
+          return;
+        }
+        auto status = batch->generate(config_.trim);
+        {
+          std::lock_guard<std::mutex> lock(status_mutex);
+          for (size_t i = 0; i < batch_indices.size(); ++i) {
+            polished[batch_indices[i]] = status[i];
+            gpu_handled[batch_indices[i]] = status[i];
+          }
+        }
+      }
+    };
+
+    std::vector<std::thread> threads;
+    threads.reserve(batches.size());
+    for (auto& b : batches) {
+      threads.emplace_back(worker, b.get());
+    }
+    for (auto& t : threads) {
+      t.join();
+    }
+
+    // CPU fallback for every window the GPU did not polish
+    // (reference cudapolisher.cpp:354-383)
+    uint64_t num_fallback = 0;
+    std::vector<bool> todo(windows_.size(), false);
+    for (uint64_t i = 0; i < windows_.size(); ++i) {
+      if (!gpu_handled[i] && windows_[i]->num_layers() >= 3) {
+        todo[i] = true;
+        ++num_fallback;
+      }
+    }
+    if (num_fallback > 0) {
+      fprintf(stderr, "[racon::HipPolisher] %lu window(s) re-polished on CPU\n",
+              static_cast<unsigned long>(num_fallback));
+      generate_consensus_cpu(polished, &todo);
+    } else {
+      logger_->log("[racon::Polisher::polish] generated consensus");
+    }
+
+    collect(dst, drop_unpolished, polished);
+  }
+
+ private:
+  std::vector<int> devices_;
+};
+
+std::unique_ptr<Polisher> createHipPolisher(std::unique_ptr<SequenceParser> sparser,
+                                            std::unique_ptr<OverlapParser> oparser,
+                                            std::unique_ptr<SequenceParser> tparser,
+                                            PolisherConfig config) {
+  return std::make_unique<HipPolisher>(std::move(sparser), std::move(oparser), std::move(tparser),
+                                       config);
+}
+
+}  // namespace rga

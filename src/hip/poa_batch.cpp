@@ -1,0 +1,311 @@
+#include "hip/poa_batch.hpp"
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstring>
+
+#include "hip/hip_common.hpp"
+
+namespace rga::hip {
+
+namespace {
+
+constexpr size_t kSeqArenaPerWindow = 96 * 1024;  // bases+weights staging per window (avg)
+
+size_t slab_bytes(const PoaLimits& L) {
+  size_t n = L.max_nodes;
+  size_t b = 0;
+  b += 4 * n;                       // letters, in_cnt, out_cnt, ring_cnt
+  b += n * L.max_edges * 8;         // in_edges(2) + in_weights(4) + out_edges(2)
+  b += n * L.max_ring * 2;          // ring
+  b += n * 2 * 4;                   // nseq, sorted, rank, work
+  b += n * 12;                      // hb_score(8) + hb_pred(4)
+  b += (2 * L.matrix_width + n) * 8;  // aln_nodes + aln_seq
+  b += (n + 1) * L.matrix_width * 2;  // matrix
+  return b;
+}
+
+}  // namespace
+
+PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch, int8_t gap,
+                   bool banded, uint32_t max_depth)
+    : device_(device), match_(match), mismatch_(mismatch), gap_(gap), max_depth_(max_depth) {
+  (void)banded;  // v1 always runs the full-width band (reference default mode)
+
+  // int16 score guard: worst |score| <= (max_nodes + matrix_width) * max|param|
+  int32_t worst = static_cast<int32_t>(limits_.max_nodes + limits_.matrix_width) *
+                  std::max({std::abs(static_cast<int32_t>(match)),
+                            std::abs(static_cast<int32_t>(mismatch)),
+                            std::abs(static_cast<int32_t>(gap))});
+  if (worst > 32000) {
+    fprintf(stderr,
+            "[rga::hip::PoaBatch] error: score parameters too large for int16 "
+            "GPU scores (|m|,|x|,|g| must keep (%u+%u)*max <= 32000)\n",
+            limits_.max_nodes, limits_.matrix_width);
+    exit(1);
+  }
+
+  RGA_HIP_CHECK(hipSetDevice(device_));
+  hipStream_t s;
+  RGA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  stream_ = s;
+
+  const size_t per_window = slab_bytes(limits_) + kSeqArenaPerWindow +
+                            limits_.max_consensus * 3 + sizeof(PoaWindowDesc) + 1024;
+  num_slabs_ = static_cast<uint32_t>(
+      std::min<size_t>(2048, std::max<size_t>(32, mem_budget / per_window)));
+  seq_arena_cap_ = static_cast<size_t>(num_slabs_) * kSeqArenaPerWindow;
+
+  const size_t max_layers = static_cast<size_t>(num_slabs_) * (max_depth_ + 1);
+
+  // ---- pinned host staging ----
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_seq_), seq_arena_cap_));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_weight_), seq_arena_cap_));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_layer_ends_), max_layers * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_layer_index_), (num_slabs_ + 1) * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_desc_),
+                              num_slabs_ * sizeof(PoaWindowDesc)));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_),
+                              static_cast<size_t>(num_slabs_) * limits_.max_consensus));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_coverage_),
+                              static_cast<size_t>(num_slabs_) * limits_.max_consensus * 2));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_len_), num_slabs_ * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_status_), num_slabs_ * 4));
+
+  // ---- device pool, carved into the arena struct ----
+  const PoaLimits& L = limits_;
+  size_t n = L.max_nodes;
+  size_t total = 0;
+  auto carve = [&total](size_t bytes) {
+    size_t off = total;
+    total += (bytes + 255) & ~size_t(255);
+    return off;
+  };
+  size_t o_seq = carve(seq_arena_cap_);
+  size_t o_wt = carve(seq_arena_cap_);
+  size_t o_ends = carve(max_layers * 4);
+  size_t o_ends_idx = carve((num_slabs_ + 1) * 4);
+  size_t o_desc = carve(num_slabs_ * sizeof(PoaWindowDesc));
+  size_t o_letters = carve(num_slabs_ * n);
+  size_t o_in_cnt = carve(num_slabs_ * n);
+  size_t o_out_cnt = carve(num_slabs_ * n);
+  size_t o_ring_cnt = carve(num_slabs_ * n);
+  size_t o_in_edges = carve(num_slabs_ * n * L.max_edges * 2);
+  size_t o_in_w = carve(num_slabs_ * n * L.max_edges * 4);
+  size_t o_out_edges = carve(num_slabs_ * n * L.max_edges * 2);
+  size_t o_ring = carve(num_slabs_ * n * L.max_ring * 2);
+  size_t o_nseq = carve(num_slabs_ * n * 2);
+  size_t o_sorted = carve(num_slabs_ * n * 2);
+  size_t o_rank = carve(num_slabs_ * n * 2);
+  size_t o_work = carve(num_slabs_ * n * 2);
+  size_t o_hb_score = carve(num_slabs_ * n * 8);
+  size_t o_hb_pred = carve(num_slabs_ * n * 4);
+  size_t o_aln_n = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
+  size_t o_aln_s = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
+  size_t o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
+  size_t o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
+  size_t o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
+  size_t o_clen = carve(num_slabs_ * 4);
+  size_t o_status = carve(num_slabs_ * 4);
+
+  RGA_HIP_CHECK(hipMalloc(&d_pool_, total));
+  auto base = static_cast<uint8_t*>(d_pool_);
+  arena_.seq_data = base + o_seq;
+  arena_.weight_data = base + o_wt;
+  arena_.layer_ends = reinterpret_cast<uint32_t*>(base + o_ends);
+  arena_.layer_ends_index = reinterpret_cast<uint32_t*>(base + o_ends_idx);
+  arena_.windows = reinterpret_cast<PoaWindowDesc*>(base + o_desc);
+  arena_.letters = base + o_letters;
+  arena_.in_cnt = base + o_in_cnt;
+  arena_.out_cnt = base + o_out_cnt;
+  arena_.ring_cnt = base + o_ring_cnt;
+  arena_.in_edges = reinterpret_cast<uint16_t*>(base + o_in_edges);
+  arena_.in_weights = reinterpret_cast<int32_t*>(base + o_in_w);
+  arena_.out_edges = reinterpret_cast<uint16_t*>(base + o_out_edges);
+  arena_.ring = reinterpret_cast<uint16_t*>(base + o_ring);
+  arena_.nseq = reinterpret_cast<uint16_t*>(base + o_nseq);
+  arena_.sorted = reinterpret_cast<uint16_t*>(base + o_sorted);
+  arena_.rank = reinterpret_cast<uint16_t*>(base + o_rank);
+  arena_.work = reinterpret_cast<uint16_t*>(base + o_work);
+  arena_.hb_score = reinterpret_cast<int64_t*>(base + o_hb_score);
+  arena_.hb_pred = reinterpret_cast<int32_t*>(base + o_hb_pred);
+  arena_.aln_nodes = reinterpret_cast<int32_t*>(base + o_aln_n);
+  arena_.aln_seq = reinterpret_cast<int32_t*>(base + o_aln_s);
+  arena_.matrix = reinterpret_cast<int16_t*>(base + o_matrix);
+  arena_.consensus = base + o_cons;
+  arena_.coverage = reinterpret_cast<uint16_t*>(base + o_cov);
+  arena_.consensus_len = reinterpret_cast<uint32_t*>(base + o_clen);
+  arena_.status = reinterpret_cast<int32_t*>(base + o_status);
+  arena_.match = match_;
+  arena_.mismatch = mismatch_;
+  arena_.gap = gap_;
+  arena_.limits = limits_;
+}
+
+PoaBatch::~PoaBatch() {
+  hipSetDevice(device_);
+  if (d_pool_ != nullptr) {
+    hipFree(d_pool_);
+  }
+  for (void* p : {static_cast<void*>(h_seq_), static_cast<void*>(h_weight_),
+                  static_cast<void*>(h_layer_ends_), static_cast<void*>(h_layer_index_),
+                  static_cast<void*>(h_desc_), static_cast<void*>(h_consensus_),
+                  static_cast<void*>(h_coverage_), static_cast<void*>(h_consensus_len_),
+                  static_cast<void*>(h_status_)}) {
+    if (p != nullptr) {
+      hipHostFree(p);
+    }
+  }
+  if (stream_ != nullptr) {
+    hipStreamDestroy(static_cast<hipStream_t>(stream_));
+  }
+}
+
+bool PoaBatch::add_window(const std::shared_ptr<Window>& window, bool* never_fits) {
+  *never_fits = false;
+  const uint32_t total_layers = window->num_layers();
+
+  // backbone must fit the DP row; otherwise this window can never run here
+  if (window->sequence(0).second + 1 > limits_.matrix_width) {
+    *never_fits = true;
+    return false;
+  }
+  if (windows_.size() >= num_slabs_) {
+    return false;
+  }
+
+  // CPU-identical layer order: backbone, then layers by window start position
+  std::vector<uint32_t> order = window->layer_order();
+
+  // count + measure what ships (skip too-long layers, cap depth)
+  uint32_t shipped = 1;
+  size_t bytes = window->sequence(0).second;
+  for (uint32_t k = 1; k < total_layers && shipped < max_depth_ + 1; ++k) {
+    uint32_t i = order[k];
+    uint32_t len = window->sequence(i).second;
+    if (len + 1 > limits_.matrix_width) {
+      continue;  // reference: exceeded_maximum_sequence_size -> skipped
+    }
+    bytes += len;
+    ++shipped;
+  }
+
+  if (seq_bytes_ + bytes > seq_arena_cap_) {
+    return false;  // arena full; try the next batch round
+  }
+
+  // pack
+  const uint32_t win_idx = static_cast<uint32_t>(windows_.size());
+  PoaWindowDesc desc;
+  desc.seq_offset = static_cast<uint32_t>(seq_bytes_);
+  desc.scratch_idx = win_idx;
+  h_layer_index_[win_idx] = static_cast<uint32_t>(num_layer_ends_);
+
+  uint32_t rel_end = 0;
+  uint32_t packed = 0;
+  for (uint32_t k = 0; k < total_layers && packed < max_depth_ + 1; ++k) {
+    uint32_t i = order[k];
+    auto seq = window->sequence(i);
+    auto qual = window->quality(i);
+    if (k > 0 && seq.second + 1 > limits_.matrix_width) {
+      continue;
+    }
+    std::memcpy(h_seq_ + seq_bytes_, seq.first, seq.second);
+    if (qual.first != nullptr) {
+      for (uint32_t b = 0; b < seq.second; ++b) {
+        h_weight_[seq_bytes_ + b] = static_cast<uint8_t>(qual.first[b]) - 33;
+      }
+    } else {
+      std::memset(h_weight_ + seq_bytes_, 1, seq.second);
+    }
+    seq_bytes_ += seq.second;
+    rel_end += seq.second;
+    h_layer_ends_[num_layer_ends_++] = rel_end;
+    ++packed;
+  }
+  desc.num_seqs = packed;
+  h_desc_[win_idx] = desc;
+
+  windows_.emplace_back(window);
+  seqs_added_.emplace_back(shipped - 1);  // layers only (effective coverage)
+  return true;
+}
+
+std::vector<bool> PoaBatch::generate(bool trim) {
+  std::vector<bool> polished(windows_.size(), false);
+  if (windows_.empty()) {
+    return polished;
+  }
+
+  RGA_HIP_CHECK(hipSetDevice(device_));
+  auto s = static_cast<hipStream_t>(stream_);
+  auto d = [&](const void* dst, const void* src, size_t bytes) {
+    RGA_HIP_CHECK(hipMemcpyAsync(const_cast<void*>(dst), src, bytes, hipMemcpyHostToDevice, s));
+  };
+  d(arena_.seq_data, h_seq_, seq_bytes_);
+  d(arena_.weight_data, h_weight_, seq_bytes_);
+  d(arena_.layer_ends, h_layer_ends_, num_layer_ends_ * 4);
+  d(arena_.layer_ends_index, h_layer_index_, windows_.size() * 4);
+  d(arena_.windows, h_desc_, windows_.size() * sizeof(PoaWindowDesc));
+
+  launch_poa_kernel(arena_, static_cast<uint32_t>(windows_.size()), stream_);
+
+  const size_t nw = windows_.size();
+  RGA_HIP_CHECK(hipMemcpyAsync(h_consensus_, arena_.consensus,
+                               nw * limits_.max_consensus, hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_coverage_, arena_.coverage,
+                               nw * limits_.max_consensus * 2, hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_consensus_len_, arena_.consensus_len, nw * 4,
+                               hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, nw * 4, hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipStreamSynchronize(s));
+
+  // CPU-parity post-processing (reference cudabatch.cpp:199-261)
+  for (size_t i = 0; i < nw; ++i) {
+    auto& window = windows_[i];
+    if (h_status_[i] != kPoaOk || h_consensus_len_[i] == 0) {
+      polished[i] = false;  // device-side failure -> CPU fallback
+      continue;
+    }
+    std::string consensus(reinterpret_cast<char*>(h_consensus_ + i * limits_.max_consensus),
+                          h_consensus_len_[i]);
+    bool status = true;
+    if (window->type() == WindowType::kTGS && trim) {
+      const uint16_t* cov = h_coverage_ + i * limits_.max_consensus;
+      uint32_t average = seqs_added_[i] / 2;
+      int32_t begin = 0, end = static_cast<int32_t>(consensus.size()) - 1;
+      for (; begin < static_cast<int32_t>(consensus.size()); ++begin) {
+        if (cov[begin] >= average) {
+          break;
+        }
+      }
+      for (; end >= 0; --end) {
+        if (cov[end] >= average) {
+          break;
+        }
+      }
+      if (begin >= end) {
+        fprintf(stderr, "[rga::hip::PoaBatch] warning: contig %lu might be chimeric in window %u!\n",
+                static_cast<unsigned long>(window->id()), window->rank());
+        status = false;
+      } else {
+        consensus = consensus.substr(begin, end - begin + 1);
+      }
+    }
+    if (status) {
+      window->set_consensus(std::move(consensus));
+    }
+    polished[i] = status;
+  }
+  return polished;
+}
+
+void PoaBatch::reset() {
+  windows_.clear();
+  seqs_added_.clear();
+  seq_bytes_ = 0;
+  num_layer_ends_ = 0;
+}
+
+}  // namespace rga::hip

@@ -1,0 +1,651 @@
+// HIP/CDNA4 POA mega-kernel: one 64-lane wavefront per window.
+//
+// Per window it loops over the layers: banded/full Needleman-Wunsch of the
+// layer against the current POA graph (rows = graph nodes in topological
+// order, columns = layer bases; the horizontal gap pass is a wave-wide
+// max-scan with slope), traceback, threading the alignment into the graph,
+// Kahn topological re-sort, and finally heaviest-bundle consensus with
+// per-base coverage. DP rows are streamed to HBM as int16 (coalesced 64-lane
+// chunks); the serial graph phases run on lane 0 and are hidden by the 16-32
+// co-resident windows per CU. No __syncthreads in the hot loop: a single
+// wavefront is synchronous by construction.
+//
+// Semantics mirror the CPU engine (src/align/poa.cpp) so GPU results are
+// deterministic and window-content-only (reference racon-gpu pins separate
+// GPU goldens; so do we — topological order here is Kahn FIFO, not spoa DFS).
+#include <hip/hip_runtime.h>
+
+#include "hip/poa_types.hpp"
+
+namespace rga::hip {
+
+namespace {
+
+constexpr int kLanes = 64;
+constexpr int32_t kNegInf = -(1 << 28);
+
+__device__ inline int32_t wave_scan_max(int32_t v, int lane) {
+  // inclusive max-scan over the 64-lane wavefront
+  for (int d = 1; d < kLanes; d <<= 1) {
+    int32_t o = __shfl_up(v, d, kLanes);
+    if (lane >= d) {
+      v = max(v, o);
+    }
+  }
+  return v;
+}
+
+struct WindowCtx {
+  // slab pointers for this window
+  uint8_t* letters;
+  uint8_t* in_cnt;
+  uint8_t* out_cnt;
+  uint8_t* ring_cnt;
+  uint16_t* in_edges;
+  int32_t* in_weights;
+  uint16_t* out_edges;
+  uint16_t* ring;
+  uint16_t* nseq;
+  uint16_t* sorted;
+  uint16_t* rank;
+  uint16_t* work;
+  int64_t* hb_score;
+  int32_t* hb_pred;
+  int32_t* aln_nodes;
+  int32_t* aln_seq;
+  int16_t* matrix;
+
+  const uint8_t* seq_base;
+  const uint8_t* weight_base;
+  const uint32_t* ends;  // layer end offsets (relative to window start)
+  uint32_t num_seqs;
+
+  uint32_t ME;  // max edges
+  uint32_t MR;  // max ring
+  uint32_t MW;  // matrix width
+  uint32_t MN;  // max nodes
+  int32_t m, x, g;
+
+  uint32_t num_nodes;
+  uint32_t seqs_in_graph;
+  int32_t status;
+};
+
+// ---------- serial (lane 0) graph helpers ----------
+
+__device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t w) {
+  uint32_t n_out = c.out_cnt[a];
+  for (uint32_t e = 0; e < n_out; ++e) {
+    if (c.out_edges[a * c.ME + e] == b) {
+      // find matching in-edge slot on b to bump the weight
+      uint32_t n_in = c.in_cnt[b];
+      for (uint32_t f = 0; f < n_in; ++f) {
+        if (c.in_edges[b * c.ME + f] == a) {
+          c.in_weights[b * c.ME + f] += w;
+          return true;
+        }
+      }
+      return true;  // unreachable for a consistent graph
+    }
+  }
+  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
+    c.status = kPoaEdgeOverflow;
+    return false;
+  }
+  c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
+  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = c.in_cnt[b];
+  c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
+  c.in_weights[b * c.ME + n_in] = w;
+  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  return true;
+}
+
+__device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
+  if (c.num_nodes >= c.MN) {
+    c.status = kPoaNodeOverflow;
+    return -1;
+  }
+  uint32_t id = c.num_nodes++;
+  c.letters[id] = letter;
+  c.in_cnt[id] = 0;
+  c.out_cnt[id] = 0;
+  c.ring_cnt[id] = 0;
+  c.nseq[id] = 0;
+  return static_cast<int32_t>(id);
+}
+
+// Threads the traceback path (stored reversed in aln_*) into the graph.
+// Mirrors Graph::add_alignment (src/align/poa.cpp).
+__device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t* wts,
+                                uint32_t len, int32_t aln_len) {
+  // first/last aligned sequence positions
+  int32_t first_pos = -1, last_pos = -1;
+  for (int32_t k = aln_len - 1; k >= 0; --k) {  // reversed storage -> forward walk
+    if (c.aln_seq[k] != -1) {
+      if (first_pos == -1) {
+        first_pos = c.aln_seq[k];
+      }
+      last_pos = c.aln_seq[k];
+    }
+  }
+
+  int32_t head = -1;
+  int32_t prev_weight = 0;
+  int32_t last_counted = -1;
+
+  auto link = [&](int32_t a, int32_t b, int32_t w) {
+    if (!add_edge_d(c, a, b, w)) {
+      return;
+    }
+    if (a != last_counted) {
+      ++c.nseq[a];
+    }
+    ++c.nseq[b];
+    last_counted = b;
+  };
+
+  if (first_pos == -1) {
+    // fully unaligned layer: append as a fresh chain
+    first_pos = len;  // head chain covers the whole sequence below
+  }
+
+  // head chain: seq[0 .. first_pos)
+  for (int32_t p = 0; p < first_pos; ++p) {
+    int32_t id = add_node_d(c, seq[p]);
+    if (id < 0) return;
+    if (head != -1) {
+      link(head, id, prev_weight + wts[p]);
+    }
+    head = id;
+    prev_weight = wts[p];
+  }
+
+  // aligned middle
+  for (int32_t k = aln_len - 1; k >= 0; --k) {
+    int32_t spos = c.aln_seq[k];
+    if (spos == -1) {
+      continue;
+    }
+    uint8_t letter = seq[spos];
+    int32_t node = c.aln_nodes[k];
+    int32_t new_id;
+    if (node == -1) {
+      new_id = add_node_d(c, letter);
+      if (new_id < 0) return;
+    } else if (c.letters[node] == letter) {
+      new_id = node;
+    } else {
+      new_id = -1;
+      uint32_t nr = c.ring_cnt[node];
+      for (uint32_t r = 0; r < nr; ++r) {
+        uint16_t aid = c.ring[node * c.MR + r];
+        if (c.letters[aid] == letter) {
+          new_id = aid;
+          break;
+        }
+      }
+      if (new_id == -1) {
+        new_id = add_node_d(c, letter);
+        if (new_id < 0) return;
+        // join the ring: new node linked with node and all its partners
+        if (nr >= c.MR) {
+          c.status = kPoaRingOverflow;
+          return;
+        }
+        for (uint32_t r = 0; r < nr; ++r) {
+          uint16_t aid = c.ring[node * c.MR + r];
+          c.ring[new_id * c.MR + r] = aid;
+          uint32_t arc = c.ring_cnt[aid];
+          if (arc >= c.MR) {
+            c.status = kPoaRingOverflow;
+            return;
+          }
+          c.ring[aid * c.MR + arc] = static_cast<uint16_t>(new_id);
+          c.ring_cnt[aid] = static_cast<uint8_t>(arc + 1);
+        }
+        c.ring[new_id * c.MR + nr] = static_cast<uint16_t>(node);
+        c.ring_cnt[new_id] = static_cast<uint8_t>(nr + 1);
+        uint32_t nrc = c.ring_cnt[node];
+        c.ring[node * c.MR + nrc] = static_cast<uint16_t>(new_id);
+        c.ring_cnt[node] = static_cast<uint8_t>(nrc + 1);
+      }
+    }
+    if (head != -1) {
+      link(head, new_id, prev_weight + wts[spos]);
+    }
+    head = new_id;
+    prev_weight = wts[spos];
+  }
+
+  // tail chain: seq[last_pos+1 .. len)
+  for (int32_t p = (last_pos == -1 ? len : last_pos + 1); p < static_cast<int32_t>(len); ++p) {
+    int32_t id = add_node_d(c, seq[p]);
+    if (id < 0) return;
+    if (head != -1) {
+      link(head, id, prev_weight + wts[p]);
+    }
+    head = id;
+    prev_weight = wts[p];
+  }
+
+  ++c.seqs_in_graph;
+}
+
+// Kahn topological sort (FIFO, deterministic). Rebuilds sorted/rank.
+__device__ void topo_sort_d(WindowCtx& c) {
+  uint32_t n = c.num_nodes;
+  for (uint32_t i = 0; i < n; ++i) {
+    c.work[i] = c.in_cnt[i];
+  }
+  uint32_t qhead = 0, qtail = 0;
+  for (uint32_t i = 0; i < n; ++i) {
+    if (c.work[i] == 0) {
+      c.sorted[qtail++] = static_cast<uint16_t>(i);
+    }
+  }
+  while (qhead < qtail) {
+    uint16_t u = c.sorted[qhead++];
+    uint32_t nout = c.out_cnt[u];
+    for (uint32_t e = 0; e < nout; ++e) {
+      uint16_t v = c.out_edges[u * c.ME + e];
+      if (--c.work[v] == 0) {
+        c.sorted[qtail++] = v;
+      }
+    }
+  }
+  for (uint32_t r = 0; r < qtail; ++r) {
+    c.rank[c.sorted[r]] = static_cast<uint16_t>(r);
+  }
+}
+
+// Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
+// Returns consensus length written into out/cov (forward order), or -1.
+__device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32_t max_out) {
+  uint32_t n = c.num_nodes;
+  for (uint32_t i = 0; i < n; ++i) {
+    c.hb_score[i] = -1;
+    c.hb_pred[i] = -1;
+  }
+
+  uint32_t max_id = c.sorted[0];
+  for (uint32_t r = 0; r < n; ++r) {
+    uint16_t nid = c.sorted[r];
+    uint32_t nin = c.in_cnt[nid];
+    for (uint32_t e = 0; e < nin; ++e) {
+      uint16_t p = c.in_edges[nid * c.ME + e];
+      int64_t w = c.in_weights[nid * c.ME + e];
+      if (c.hb_score[nid] < w ||
+          (c.hb_score[nid] == w && c.hb_pred[nid] != -1 &&
+           c.hb_score[c.hb_pred[nid]] <= c.hb_score[p])) {
+        c.hb_score[nid] = w;
+        c.hb_pred[nid] = p;
+      }
+    }
+    if (c.hb_pred[nid] != -1) {
+      c.hb_score[nid] += c.hb_score[c.hb_pred[nid]];
+    }
+    if (c.hb_score[max_id] < c.hb_score[nid]) {
+      max_id = nid;
+    }
+  }
+
+  // branch completion until we end on a sink (bounded: the restart rank is
+  // strictly increasing, so at most n rounds; all-zero-weight graphs would
+  // otherwise spin forever — those windows fail over to the CPU instead)
+  uint32_t guard = 0;
+  uint32_t prev_rank = 0;
+  while (c.out_cnt[max_id] != 0) {
+    if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
+      c.status = kPoaConsensusOverflow;
+      return -1;
+    }
+    prev_rank = c.rank[max_id];
+    uint32_t rank0 = c.rank[max_id];
+    // invalidate alternative branches
+    uint32_t nout = c.out_cnt[max_id];
+    for (uint32_t e = 0; e < nout; ++e) {
+      uint16_t endn = c.out_edges[max_id * c.ME + e];
+      uint32_t nin = c.in_cnt[endn];
+      for (uint32_t f = 0; f < nin; ++f) {
+        uint16_t o = c.in_edges[endn * c.ME + f];
+        if (o != max_id) {
+          c.hb_score[o] = -1;
+        }
+      }
+    }
+    int64_t best = 0;
+    uint32_t best_id = 0;
+    for (uint32_t r = rank0 + 1; r < n; ++r) {
+      uint16_t nid = c.sorted[r];
+      c.hb_score[nid] = -1;
+      c.hb_pred[nid] = -1;
+      uint32_t nin = c.in_cnt[nid];
+      for (uint32_t e = 0; e < nin; ++e) {
+        uint16_t p = c.in_edges[nid * c.ME + e];
+        if (c.hb_score[p] == -1) {
+          continue;
+        }
+        int64_t w = c.in_weights[nid * c.ME + e];
+        if (c.hb_score[nid] < w ||
+            (c.hb_score[nid] == w && c.hb_pred[nid] != -1 &&
+             c.hb_score[c.hb_pred[nid]] <= c.hb_score[p])) {
+          c.hb_score[nid] = w;
+          c.hb_pred[nid] = p;
+        }
+      }
+      if (c.hb_pred[nid] != -1) {
+        c.hb_score[nid] += c.hb_score[c.hb_pred[nid]];
+      }
+      if (best < c.hb_score[nid]) {
+        best = c.hb_score[nid];
+        best_id = nid;
+      }
+    }
+    max_id = best_id;
+  }
+
+  // backtrack: path length first
+  int32_t path_len = 0;
+  int32_t idw = static_cast<int32_t>(max_id);
+  while (idw != -1) {
+    ++path_len;
+    idw = c.hb_pred[idw];
+  }
+  if (path_len > static_cast<int32_t>(max_out)) {
+    c.status = kPoaConsensusOverflow;
+    return -1;
+  }
+  idw = static_cast<int32_t>(max_id);
+  for (int32_t k = path_len - 1; k >= 0; --k) {
+    uint32_t node = static_cast<uint32_t>(idw);
+    out[k] = c.letters[node];
+    uint32_t covv = c.nseq[node];
+    uint32_t nr = c.ring_cnt[node];
+    for (uint32_t r = 0; r < nr; ++r) {
+      covv += c.nseq[c.ring[node * c.MR + r]];
+    }
+    cov[k] = static_cast<uint16_t>(min(covv, 65535u));
+    idw = c.hb_pred[idw];
+  }
+  return path_len;
+}
+
+// ---------- the mega-kernel ----------
+
+__launch_bounds__(kLanes, 8)
+__global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
+  const uint32_t win = blockIdx.x;
+  if (win >= num_windows) {
+    return;
+  }
+  const int lane = threadIdx.x;
+  const PoaWindowDesc desc = a.windows[win];
+  const PoaLimits L = a.limits;
+  const uint32_t slab = desc.scratch_idx;
+
+  __shared__ uint8_t lds_seq[1024];
+
+  WindowCtx c;
+  c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
+  c.in_cnt = a.in_cnt + static_cast<size_t>(slab) * L.max_nodes;
+  c.out_cnt = a.out_cnt + static_cast<size_t>(slab) * L.max_nodes;
+  c.ring_cnt = a.ring_cnt + static_cast<size_t>(slab) * L.max_nodes;
+  c.in_edges = a.in_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
+  c.in_weights = a.in_weights + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
+  c.out_edges = a.out_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
+  c.ring = a.ring + static_cast<size_t>(slab) * L.max_nodes * L.max_ring;
+  c.nseq = a.nseq + static_cast<size_t>(slab) * L.max_nodes;
+  c.sorted = a.sorted + static_cast<size_t>(slab) * L.max_nodes;
+  c.rank = a.rank + static_cast<size_t>(slab) * L.max_nodes;
+  c.work = a.work + static_cast<size_t>(slab) * L.max_nodes;
+  c.hb_score = a.hb_score + static_cast<size_t>(slab) * L.max_nodes;
+  c.hb_pred = a.hb_pred + static_cast<size_t>(slab) * L.max_nodes;
+  c.aln_nodes = a.aln_nodes + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
+  c.aln_seq = a.aln_seq + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
+  c.matrix = a.matrix + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
+
+  c.seq_base = a.seq_data + desc.seq_offset;
+  c.weight_base = a.weight_data + desc.seq_offset;
+  c.ends = a.layer_ends + a.layer_ends_index[win];
+  c.num_seqs = desc.num_seqs;
+  c.ME = L.max_edges;
+  c.MR = L.max_ring;
+  c.MW = L.matrix_width;
+  c.MN = L.max_nodes;
+  c.m = a.match;
+  c.x = a.mismatch;
+  c.g = a.gap;
+  c.status = kPoaOk;
+
+  // ---- init graph from the backbone (layer 0), lane-parallel ----
+  const uint32_t bb_len = c.ends[0];
+  const uint8_t* bb_seq = c.seq_base;
+  const uint8_t* bb_wts = c.weight_base;
+  for (uint32_t i = lane; i < bb_len; i += kLanes) {
+    c.letters[i] = bb_seq[i];
+    c.ring_cnt[i] = 0;
+    c.nseq[i] = bb_len >= 2 ? 1 : 0;
+    c.sorted[i] = static_cast<uint16_t>(i);
+    c.rank[i] = static_cast<uint16_t>(i);
+    if (i == 0) {
+      c.in_cnt[i] = 0;
+    } else {
+      c.in_cnt[i] = 1;
+      c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
+      c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
+    }
+    if (i + 1 < bb_len) {
+      c.out_cnt[i] = 1;
+      c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
+    } else {
+      c.out_cnt[i] = 0;
+    }
+  }
+  c.num_nodes = bb_len;
+  c.seqs_in_graph = 1;
+  __threadfence_block();  // backbone graph writes -> visible to all lanes
+
+  // ---- per-layer loop ----
+  for (uint32_t layer = 1; layer < c.num_seqs && c.status == kPoaOk; ++layer) {
+    const uint32_t beg = c.ends[layer - 1];
+    const uint32_t len = c.ends[layer] - beg;
+    const uint8_t* seq = c.seq_base + beg;
+    const uint8_t* wts = c.weight_base + beg;
+    if (len == 0 || len + 1 > c.MW) {
+      continue;  // host should have filtered; skip defensively
+    }
+
+    for (uint32_t j = lane; j < len; j += kLanes) {
+      lds_seq[j] = seq[j];
+    }
+
+    const uint32_t n = c.num_nodes;
+    const uint32_t width = len + 1;
+    const uint32_t chunks = (len + kLanes - 1) / kLanes;
+
+    // row 0: all-gap prefix of the layer
+    for (uint32_t j = lane; j < width; j += kLanes) {
+      c.matrix[j] = static_cast<int16_t>(static_cast<int32_t>(j) * c.g);
+    }
+    __threadfence_block();
+
+    int32_t best_score = kNegInf;
+    uint32_t best_row = 0;
+
+    for (uint32_t r = 0; r < n; ++r) {
+      const uint16_t node = c.sorted[r];
+      const uint8_t letter = c.letters[node];
+      const uint32_t nin = c.in_cnt[node];
+      int16_t* Hrow = c.matrix + static_cast<size_t>(r + 1) * c.MW;
+
+      // first column (j = 0): max over preds of Hp[0] + gap
+      int32_t h0;
+      {
+        int32_t best0 = kNegInf;
+        if (nin == 0) {
+          best0 = 0;
+        } else {
+          for (uint32_t e = 0; e < nin; ++e) {
+            uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
+            best0 = max(best0, static_cast<int32_t>(c.matrix[static_cast<size_t>(p) * c.MW]));
+          }
+        }
+        h0 = best0 + c.g;
+        if (lane == 0) {
+          Hrow[0] = static_cast<int16_t>(h0);
+        }
+      }
+
+      // carry for the horizontal scan, in u-space: u(j) = H[j] - j*g
+      int32_t carry_u = h0;
+      int32_t last_col_val = kNegInf;
+
+      for (uint32_t k = 0; k < chunks; ++k) {
+        const uint32_t j = 1 + k * kLanes + lane;  // column this lane owns
+        int32_t v = kNegInf;
+        if (j < width) {
+          const int32_t sub = (lds_seq[j - 1] == letter) ? c.m : c.x;
+          if (nin == 0) {
+            const int16_t* Hp = c.matrix;  // row 0
+            v = max(static_cast<int32_t>(Hp[j - 1]) + sub, static_cast<int32_t>(Hp[j]) + c.g);
+          } else {
+            for (uint32_t e = 0; e < nin; ++e) {
+              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
+              const int16_t* Hp = c.matrix + static_cast<size_t>(p) * c.MW;
+              v = max(v, max(static_cast<int32_t>(Hp[j - 1]) + sub,
+                             static_cast<int32_t>(Hp[j]) + c.g));
+            }
+          }
+        }
+        // horizontal pass: H[j] = max(v[j], H[j-1] + g) as a max-scan in u-space
+        int32_t u = (j < width) ? v - static_cast<int32_t>(j) * c.g : kNegInf;
+        u = wave_scan_max(u, lane);
+        u = max(u, carry_u);
+        const int32_t h = u + static_cast<int32_t>(j) * c.g;
+        if (j < width) {
+          Hrow[j] = static_cast<int16_t>(h);
+          if (j == len) {
+            last_col_val = h;
+          }
+        }
+        carry_u = __shfl(u, kLanes - 1, kLanes);
+      }
+
+      // this row's stores must be visible to every lane before the next row
+      // (cross-lane read-after-write through global memory within one wave)
+      __threadfence_block();
+
+      // end-node max (strict >, first in topological order wins)
+      if (c.out_cnt[node] == 0) {
+        const int src_lane = static_cast<int>((len - 1) % kLanes);
+        const int32_t lc = __shfl(last_col_val, src_lane, kLanes);
+        if (lc > best_score) {
+          best_score = lc;
+          best_row = r + 1;
+        }
+      }
+    }
+
+    // ---- serial phases on lane 0 ----
+    if (lane == 0) {
+      // traceback (priority: diagonal preds in order, vertical, horizontal)
+      int32_t aln_len = 0;
+      uint32_t i = best_row, j = len;
+      while (!(i == 0 && j == 0)) {
+        const int32_t H_ij = c.matrix[static_cast<size_t>(i) * c.MW + j];
+        uint32_t prev_i = i, prev_j = j;
+        bool found = false;
+        if (i != 0 && j != 0) {
+          const uint16_t node = c.sorted[i - 1];
+          const int32_t mc = (lds_seq[j - 1] == c.letters[node]) ? c.m : c.x;
+          const uint32_t nin = c.in_cnt[node];
+          if (nin == 0) {
+            if (H_ij == c.matrix[j - 1] + mc) {
+              prev_i = 0;
+              prev_j = j - 1;
+              found = true;
+            }
+          } else {
+            for (uint32_t e = 0; e < nin && !found; ++e) {
+              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
+              if (H_ij == c.matrix[static_cast<size_t>(p) * c.MW + j - 1] + mc) {
+                prev_i = p;
+                prev_j = j - 1;
+                found = true;
+              }
+            }
+          }
+        }
+        if (!found && i != 0) {
+          const uint16_t node = c.sorted[i - 1];
+          const uint32_t nin = c.in_cnt[node];
+          if (nin == 0) {
+            if (H_ij == c.matrix[j] + c.g) {
+              prev_i = 0;
+              found = true;
+            }
+          } else {
+            for (uint32_t e = 0; e < nin && !found; ++e) {
+              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
+              if (H_ij == c.matrix[static_cast<size_t>(p) * c.MW + j] + c.g) {
+                prev_i = p;
+                found = true;
+              }
+            }
+          }
+        }
+        if (!found && j != 0) {
+          if (H_ij == c.matrix[static_cast<size_t>(i) * c.MW + j - 1] + c.g) {
+            prev_j = j - 1;
+            found = true;
+          }
+        }
+        if (!found) {
+          c.status = kPoaConsensusOverflow;  // inconsistent DP: fail window
+          break;
+        }
+        c.aln_nodes[aln_len] = (i == prev_i) ? -1 : static_cast<int32_t>(c.sorted[i - 1]);
+        c.aln_seq[aln_len] = (j == prev_j) ? -1 : static_cast<int32_t>(j - 1);
+        ++aln_len;
+        i = prev_i;
+        j = prev_j;
+      }
+
+      if (c.status == kPoaOk) {
+        add_alignment_d(c, seq, wts, len, aln_len);
+      }
+      if (c.status == kPoaOk) {
+        topo_sort_d(c);
+      }
+    }
+
+    // lane 0's graph updates must be visible to the whole wave
+    __threadfence_block();
+    // broadcast updated scalars from lane 0 to the wave
+    c.num_nodes = __shfl(c.num_nodes, 0, kLanes);
+    c.seqs_in_graph = __shfl(c.seqs_in_graph, 0, kLanes);
+    c.status = __shfl(c.status, 0, kLanes);
+  }
+
+  // ---- consensus ----
+  if (lane == 0) {
+    uint8_t* out = a.consensus + static_cast<size_t>(win) * L.max_consensus;
+    uint16_t* cov = a.coverage + static_cast<size_t>(win) * L.max_consensus;
+    int32_t clen = -1;
+    if (c.status == kPoaOk) {
+      clen = consensus_d(c, out, cov, L.max_consensus);
+    }
+    a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
+    a.status[win] = c.status;
+  }
+}
+
+}  // namespace
+
+void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t num_windows, void* stream) {
+  hipLaunchKernelGGL(poa_window_kernel, dim3(num_windows), dim3(kLanes), 0,
+                     static_cast<hipStream_t>(stream), arena, num_windows);
+}
+
+}  // namespace rga::hip

@@ -1,0 +1,85 @@
+// Device-side batch layout for the HIP POA engine.
+//
+// Capability parity target: GenomeWorks cudapoa as driven by reference
+// src/cuda/cudabatch.cpp (BatchConfig(1023, 200, 256, ...), add_poa_group /
+// generate_poa / get_consensus contract) — re-designed for CDNA4: one 64-lane
+// wavefront per window, full-width DP rows streamed through HBM as int16,
+// serial graph phases on lane 0 hidden by 16-32 co-resident windows per CU.
+#pragma once
+
+#include <cstdint>
+
+namespace rga::hip {
+
+// Capacity model (per window). Windows exceeding any cap are rejected or
+// failed on device and fall back to the CPU path (reference contract:
+// cudapolisher.cpp:354-383).
+struct PoaLimits {
+  uint32_t max_seq_len = 1023;    // bases per layer (incl. backbone)
+  uint32_t max_depth = 200;       // layers per window (MAX_DEPTH_PER_WINDOW)
+  uint32_t max_nodes = 2047;      // POA graph nodes per window
+  uint32_t max_edges = 48;        // in- or out-edges per node
+  uint32_t max_ring = 4;          // aligned-ring partners per node
+  uint32_t matrix_width = 1024;   // DP row width (max_seq_len + 1)
+  uint32_t max_consensus = 2048;  // consensus output cap
+};
+
+// Window status codes produced by the kernel.
+enum PoaStatus : int32_t {
+  kPoaOk = 0,
+  kPoaNodeOverflow = 1,
+  kPoaEdgeOverflow = 2,
+  kPoaRingOverflow = 3,
+  kPoaConsensusOverflow = 4,
+  kPoaNotRun = 5,
+};
+
+// Per-window input descriptor (layers already sorted: backbone first, then
+// by window start position — the same order as the CPU path).
+struct PoaWindowDesc {
+  uint32_t seq_offset;   // byte offset into the packed seq/weight arena
+  uint32_t num_seqs;     // layers shipped to the device (<= max_depth + 1)
+  uint32_t scratch_idx;  // which device slab this window uses
+};
+
+// Device arena pointers (one allocation, carved into slabs).
+struct PoaDeviceArena {
+  // inputs (packed tight, H2D once per batch)
+  const uint8_t* seq_data;     // concatenated layer bases
+  const uint8_t* weight_data;  // matching per-base weights
+  const uint32_t* layer_ends;  // per layer: end offset within window's span
+  const uint32_t* layer_ends_index;  // per window: first index into layer_ends
+  const PoaWindowDesc* windows;
+
+  // per-window graph slabs (device scratch, indexed by scratch_idx)
+  uint8_t* letters;      // [max_nodes]
+  uint8_t* in_cnt;       // [max_nodes]
+  uint8_t* out_cnt;      // [max_nodes]
+  uint8_t* ring_cnt;     // [max_nodes]
+  uint16_t* in_edges;    // [max_nodes * max_edges]
+  int32_t* in_weights;   // [max_nodes * max_edges]
+  uint16_t* out_edges;   // [max_nodes * max_edges]
+  uint16_t* ring;        // [max_nodes * max_ring]
+  uint16_t* nseq;        // [max_nodes] sequences touching node (coverage)
+  uint16_t* sorted;      // [max_nodes] topological order
+  uint16_t* rank;        // [max_nodes] node -> rank
+  uint16_t* work;        // [max_nodes] Kahn queue / in-degree scratch
+  int64_t* hb_score;     // [max_nodes] heaviest-bundle running scores
+  int32_t* hb_pred;      // [max_nodes]
+  int32_t* aln_nodes;    // [2 * matrix_width + max_nodes] alignment node ids
+  int32_t* aln_seq;      // [same] alignment sequence positions
+  int16_t* matrix;       // [(max_nodes + 1) * matrix_width] DP scores
+
+  // outputs (D2H once per batch)
+  uint8_t* consensus;     // [max_consensus] per window, reversed on host
+  uint16_t* coverage;     // [max_consensus] per window
+  uint32_t* consensus_len;  // [1] per window
+  int32_t* status;          // [1] per window
+
+  int8_t match, mismatch, gap;
+  PoaLimits limits;
+};
+
+void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t num_windows, void* stream);
+
+}  // namespace rga::hip

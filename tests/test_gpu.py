@@ -1,0 +1,64 @@
+"""GPU (MI355X) tests: HIP POA pipeline numerics vs the CPU engine,
+determinism, and quality vs truth. Run via gpurun: pytest tests -m gpu."""
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def require_gpu(racon):
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU on this host")
+
+
+def test_gpu_polish_runs_and_improves_draft(racon, sample, fasta_reader):
+    truth = list(fasta_reader(sample["reference"]).values())[0]
+    draft = list(fasta_reader(sample["layout"]).values())[0]
+    before = racon.edit_distance(draft, truth)
+
+    out = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                       threads=4, poa_batches=1)
+    assert len(out) == 1
+    after = racon.edit_distance(out[0][1], truth)
+    assert after < before * 0.2, (before, after)
+
+
+def test_gpu_matches_cpu_closely(racon, sample):
+    cpu = racon.polish(sample["reads"], sample["overlaps"], sample["layout"], threads=4)
+    gpu = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                       threads=4, poa_batches=1)
+    assert len(cpu) == len(gpu)
+    # GPU tie-breaking (Kahn topological order) may differ from the CPU DFS
+    # order; the reference pins different goldens for each path too. Demand
+    # near-identity: < 0.5% divergence.
+    ed = racon.edit_distance(cpu[0][1], gpu[0][1])
+    assert ed < 0.005 * len(cpu[0][1]), ed
+
+
+def test_gpu_deterministic(racon, sample):
+    a = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                     threads=4, poa_batches=1)
+    b = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                     threads=2, poa_batches=2)
+    # window consensus must depend only on window content, not batch layout
+    assert a == b
+
+
+def test_gpu_reference_sample_golden(racon, ref_data, fasta_reader):
+    """Pinned GPU golden on the reference lambda-phage sample (our HIP path;
+    the reference pins its own different CUDA goldens, racon_test.cpp:312)."""
+    ref = list(fasta_reader(str(ref_data / "sample_reference.fasta.gz")).values())[0].upper()
+    out = racon.polish(str(ref_data / "sample_reads.fastq.gz"),
+                       str(ref_data / "sample_overlaps.paf.gz"),
+                       str(ref_data / "sample_layout.fasta.gz"),
+                       threads=4, match=5, mismatch=-4, gap=-8, poa_batches=1)
+    assert len(out) == 1
+    rc = racon.reverse_complement(out[0][1])
+    ed = racon.edit_distance(rc, ref)
+    # CPU path: 1314; reference CPU golden 1312, reference CUDA golden 1385.
+    # Bound the HIP path to the same quality band; the exact value is pinned
+    # once measured on hardware.
+    assert ed < 1500, ed
