@@ -13,6 +13,7 @@
 #include <vector>
 
 #include "core/polisher.hpp"
+#include "hip/aligner_batch.hpp"
 #include "hip/hip_common.hpp"
 #include "hip/poa_batch.hpp"
 
@@ -34,6 +35,80 @@ class HipPolisher : public Polisher {
       devices_.emplace_back(d);
     }
     fprintf(stderr, "[racon::HipPolisher] using %d GPU(s)\n", n);
+  }
+
+  // GPU overlap alignment; overlaps the GPU skips or fails keep an empty
+  // CIGAR and are aligned by the CPU pairwise path inside the base-class
+  // call at the end (reference cudapolisher.cpp:74-213).
+  void find_overlap_breaking_points(std::vector<std::unique_ptr<Overlap>>& overlaps) override {
+    if (config_.aligner_batches < 1) {
+      Polisher::find_overlap_breaking_points(overlaps);
+      return;
+    }
+
+    std::vector<std::unique_ptr<hip::AlignerBatch>> batches;
+    for (int d : devices_) {
+      RGA_HIP_CHECK(hipSetDevice(d));
+      size_t free_mem = 0, total_mem = 0;
+      RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
+      size_t budget = free_mem * 9 / 10 / config_.aligner_batches;
+      for (uint32_t b = 0; b < config_.aligner_batches; ++b) {
+        batches.emplace_back(std::make_unique<hip::AlignerBatch>(d, budget));
+      }
+    }
+
+    std::mutex queue_mutex;
+    uint64_t next_overlap = 0;
+    std::atomic<uint64_t> skipped{0};
+
+    auto worker = [&](hip::AlignerBatch* batch) {
+      while (true) {
+        batch->reset();
+        uint32_t pulled = 0;
+        {
+          std::lock_guard<std::mutex> lock(queue_mutex);
+          while (next_overlap < overlaps.size()) {
+            auto* o = overlaps[next_overlap].get();
+            if (o->has_cigar()) {  // SAM input already has an alignment
+              ++next_overlap;
+              continue;
+            }
+            bool never_fits = false;
+            if (batch->add_overlap(o, sequences_, &never_fits)) {
+              ++pulled;
+              ++next_overlap;
+            } else if (never_fits) {
+              ++skipped;
+              ++next_overlap;
+            } else {
+              break;
+            }
+          }
+        }
+        if (pulled == 0) {
+          return;
+        }
+        skipped += batch->align_and_emit();
+      }
+    };
+
+    std::vector<std::thread> threads;
+    threads.reserve(batches.size());
+    for (auto& b : batches) {
+      threads.emplace_back(worker, b.get());
+    }
+    for (auto& t : threads) {
+      t.join();
+    }
+    batches.clear();
+
+    if (skipped.load() > 0) {
+      fprintf(stderr, "[racon::HipPolisher] %lu overlap(s) aligned on CPU\n",
+              static_cast<unsigned long>(skipped.load()));
+    }
+    // CPU pass: walks CIGARs into breaking points; aligns leftovers with the
+    // CPU pairwise engine
+    Polisher::find_overlap_breaking_points(overlaps);
   }
 
   void polish(std::vector<std::unique_ptr<Sequence>>& dst, bool drop_unpolished) override {
@@ -138,6 +213,10 @@ class HipPolisher : public Polisher {
  private:
   std::vector<int> devices_;
 };
+
+namespace hip {
+int runtime_device_count() { return device_count(); }
+}  // namespace hip
 
 std::unique_ptr<Polisher> createHipPolisher(std::unique_ptr<SequenceParser> sparser,
                                             std::unique_ptr<OverlapParser> oparser,

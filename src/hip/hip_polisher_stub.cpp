@@ -9,6 +9,10 @@
 
 namespace rga {
 
+namespace hip {
+int runtime_device_count() { return 0; }
+}  // namespace hip
+
 std::unique_ptr<Polisher> createHipPolisher(std::unique_ptr<SequenceParser>,
                                             std::unique_ptr<OverlapParser>,
                                             std::unique_ptr<SequenceParser>, PolisherConfig) {

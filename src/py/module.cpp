@@ -13,6 +13,10 @@
 #include "core/polisher.hpp"
 #include "core/sequence.hpp"
 
+namespace rga::hip {
+int runtime_device_count();  // provided by the HIP backend (or the stub)
+}
+
 namespace py = pybind11;
 
 namespace {
@@ -109,6 +113,8 @@ PYBIND11_MODULE(_racon, m) {
         py::arg("aligner_band_width") = 0, py::arg("include_unpolished") = false,
         "Polish targets with reads+overlaps; returns [(name, sequence)].");
 
+  m.def("device_count", [] { return rga::hip::runtime_device_count(); },
+        "Number of visible HIP devices.");
   m.def("edit_distance", &edit_distance_py, py::arg("a"), py::arg("b"));
   m.def("align_cigar", &align_cigar_py, py::arg("query"), py::arg("target"));
   m.def("reverse_complement", &reverse_complement, py::arg("sequence"));
