@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""racon_wrapper: subsampling + target-splitting front-end for racon.
+
+Capability parity with the reference wrapper
+(/root/reference/scripts/racon_wrapper.py) — same CLI, same rampler-driven
+--subsample / --split preprocessing, sequential racon runs per target chunk
+with concatenated stdout. Two deliberate improvements over the reference:
+
+* GPU flags (-c/--cudapoa-batches, --cudaaligner-batches,
+  -b/--cuda-banded-alignment, --cudaaligner-band-width) are ACTUALLY
+  forwarded to racon; the reference parses but drops them
+  (racon_wrapper.py:38-40,130-131 are commented out there).
+* binaries are located at runtime (RACON_BIN / RAMPLER_BIN env vars, then
+  the in-tree build/ directory, then PATH) instead of CMake path
+  substitution, so the wrapper runs uninstalled.
+"""
+
+import argparse
+import os
+import shutil
+import subprocess
+import sys
+import tempfile
+
+
+def eprint(*args, **kwargs):
+    print(*args, file=sys.stderr, flush=True, **kwargs)
+
+
+def find_binary(env_var, name):
+    path = os.environ.get(env_var)
+    if path and os.path.isfile(path):
+        return path
+    here = os.path.dirname(os.path.abspath(__file__))
+    candidate = os.path.join(here, "..", "build", name)
+    if os.path.isfile(candidate):
+        return candidate
+    found = shutil.which(name)
+    if found:
+        return found
+    eprint(f"[racon_wrapper] error: unable to locate the {name} binary "
+           f"(set {env_var} or build the project)")
+    sys.exit(1)
+
+
+def sequence_extension(path):
+    fasta_exts = (".fasta", ".fasta.gz", ".fa", ".fa.gz")
+    return ".fasta" if path.endswith(fasta_exts) else ".fastq"
+
+
+def main():
+    parser = argparse.ArgumentParser(
+        description="racon front-end adding read subsampling (lower runtime) "
+                    "and target splitting (lower memory); the racon CLI is "
+                    "otherwise passed through unchanged.",
+        formatter_class=argparse.ArgumentDefaultsHelpFormatter)
+    parser.add_argument("sequences", help="FASTA/FASTQ (may be gzipped) reads")
+    parser.add_argument("overlaps", help="MHAP/PAF/SAM (may be gzipped) overlaps")
+    parser.add_argument("target_sequences", help="FASTA/FASTQ (may be gzipped) targets")
+    parser.add_argument("--split", help="split target sequences into chunks of this many bytes")
+    parser.add_argument("--subsample", nargs=2, metavar=("REF_LEN", "COV"),
+                        help="subsample reads to COV x coverage of a REF_LEN bp reference")
+    parser.add_argument("-u", "--include-unpolished", action="store_true")
+    parser.add_argument("-f", "--fragment-correction", action="store_true")
+    parser.add_argument("-w", "--window-length", default=500)
+    parser.add_argument("-q", "--quality-threshold", default=10.0)
+    parser.add_argument("-e", "--error-threshold", default=0.3)
+    parser.add_argument("--no-trimming", action="store_true")
+    parser.add_argument("-m", "--match", default=5)
+    parser.add_argument("-x", "--mismatch", default=-4)
+    parser.add_argument("-g", "--gap", default=-8)
+    parser.add_argument("-t", "--threads", default=1)
+    parser.add_argument("-c", "--cudapoa-batches", default=0,
+                        help="number of HIP POA batches per GPU")
+    parser.add_argument("--cudaaligner-batches", default=0,
+                        help="number of HIP aligner batches per GPU")
+    parser.add_argument("-b", "--cuda-banded-alignment", action="store_true",
+                        help="banded approximation for GPU POA alignment")
+    parser.add_argument("--cudaaligner-band-width", default=0,
+                        help="band width for GPU alignment (0 = auto)")
+    args = parser.parse_args()
+
+    racon = find_binary("RACON_BIN", "racon")
+    rampler = find_binary("RAMPLER_BIN", "rampler")
+
+    sequences = os.path.abspath(args.sequences)
+    overlaps = os.path.abspath(args.overlaps)
+    targets = os.path.abspath(args.target_sequences)
+
+    work = tempfile.mkdtemp(prefix="racon_work_directory_", dir=os.getcwd())
+    try:
+        if args.subsample is not None:
+            ref_len, cov = args.subsample
+            eprint("[racon_wrapper] subsampling reads with rampler")
+            subprocess.run([rampler, "-o", work, "subsample", sequences, ref_len, cov],
+                           check=True)
+            base = os.path.basename(sequences).split(".")[0]
+            sequences = os.path.join(work, f"{base}_{cov}x{sequence_extension(sequences)}")
+            if not os.path.isfile(sequences):
+                eprint("[racon_wrapper] error: subsampled sequences not found")
+                sys.exit(1)
+
+        if args.split is not None:
+            eprint("[racon_wrapper] splitting targets with rampler")
+            subprocess.run([rampler, "-o", work, "split", targets, str(args.split)],
+                           check=True)
+            base = os.path.basename(targets).split(".")[0]
+            ext = sequence_extension(targets)
+            chunks = []
+            i = 0
+            while os.path.isfile(os.path.join(work, f"{base}_{i}{ext}")):
+                chunks.append(os.path.join(work, f"{base}_{i}{ext}"))
+                i += 1
+            eprint(f"[racon_wrapper] total number of splits: {i}")
+            if not chunks:
+                eprint("[racon_wrapper] error: split target sequences not found")
+                sys.exit(1)
+        else:
+            chunks = [targets]
+
+        cmd = [racon]
+        if args.include_unpolished:
+            cmd.append("-u")
+        if args.fragment_correction:
+            cmd.append("-f")
+        if args.no_trimming:
+            cmd.append("--no-trimming")
+        if args.cuda_banded_alignment:
+            cmd.append("-b")
+        cmd += ["-w", str(args.window_length), "-q", str(args.quality_threshold),
+                "-e", str(args.error_threshold), "-m", str(args.match),
+                "-x", str(args.mismatch), "-g", str(args.gap), "-t", str(args.threads)]
+        if int(args.cudapoa_batches):
+            cmd += ["-c", str(args.cudapoa_batches)]
+        if int(args.cudaaligner_batches):
+            cmd += ["--cudaaligner-batches", str(args.cudaaligner_batches)]
+        if int(args.cudaaligner_band_width):
+            cmd += ["--cudaaligner-band-width", str(args.cudaaligner_band_width)]
+        cmd += [sequences, overlaps, None]
+
+        for chunk in chunks:
+            eprint(f"[racon_wrapper] polishing {os.path.basename(chunk)}")
+            cmd[-1] = chunk
+            subprocess.run(cmd, check=True)
+    except subprocess.CalledProcessError as e:
+        eprint(f"[racon_wrapper] error: subprocess failed: {e}")
+        sys.exit(1)
+    finally:
+        shutil.rmtree(work, ignore_errors=True)
+
+
+if __name__ == "__main__":
+    main()

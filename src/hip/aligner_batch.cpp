@@ -98,11 +98,9 @@ bool AlignerBatch::add_overlap(Overlap* overlap,
   AlnDesc d;
   d.q_offset = static_cast<uint32_t>(seq_bytes_);
   d.q_len = q.second;
-  std::memcpy(h_seqs_ + seq_bytes_, q.first, q.second);
   seq_bytes_ += q.second;
   d.t_offset = static_cast<uint32_t>(seq_bytes_);
   d.t_len = t.second;
-  std::memcpy(h_seqs_ + seq_bytes_, t.first, t.second);
   seq_bytes_ += t.second;
   d.moves_offset = moves_dw_;
   moves_dw_ += mdw;
@@ -111,13 +109,24 @@ bool AlignerBatch::add_overlap(Overlap* overlap,
 
   h_descs_[overlaps_.size()] = d;
   overlaps_.emplace_back(overlap);
+  pending_.push_back({q.first, t.first});
   return true;
+}
+
+void AlignerBatch::pack() {
+  for (; packed_upto_ < overlaps_.size(); ++packed_upto_) {
+    const AlnDesc& d = h_descs_[packed_upto_];
+    const PendingSpan& p = pending_[packed_upto_];
+    std::memcpy(h_seqs_ + d.q_offset, p.q, d.q_len);
+    std::memcpy(h_seqs_ + d.t_offset, p.t, d.t_len);
+  }
 }
 
 uint32_t AlignerBatch::align_and_emit() {
   if (overlaps_.empty()) {
     return 0;
   }
+  pack();
   RGA_HIP_CHECK(hipSetDevice(device_));
   auto s = static_cast<hipStream_t>(stream_);
   RGA_HIP_CHECK(hipMemcpyAsync(const_cast<uint8_t*>(arena_.seqs), h_seqs_, seq_bytes_,
@@ -171,6 +180,8 @@ uint32_t AlignerBatch::align_and_emit() {
 
 void AlignerBatch::reset() {
   overlaps_.clear();
+  pending_.clear();
+  packed_upto_ = 0;
   seq_bytes_ = 0;
   moves_dw_ = 0;
   path_bytes_ = 0;

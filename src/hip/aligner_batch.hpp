@@ -22,10 +22,17 @@ class AlignerBatch {
   AlignerBatch(const AlignerBatch&) = delete;
   AlignerBatch& operator=(const AlignerBatch&) = delete;
 
-  // Packs the overlap's query/target spans. Returns false when the batch is
-  // full; never_fits is set when this overlap cannot run on the GPU at all.
+  // Reserves arena space for the overlap's query/target spans (bookkeeping
+  // only — callable under the shared queue lock). Returns false when the
+  // batch is full; never_fits is set when this overlap cannot run on the GPU
+  // at all. The actual copies happen in pack().
   bool add_overlap(Overlap* overlap, const std::vector<std::unique_ptr<Sequence>>& sequences,
                    bool* never_fits);
+
+  // Copies every reserved span into the pinned staging buffers. Called
+  // outside the queue lock so fills from different batches proceed in
+  // parallel.
+  void pack();
 
   uint32_t size() const { return static_cast<uint32_t>(overlaps_.size()); }
 
@@ -56,6 +63,12 @@ class AlignerBatch {
   size_t moves_dw_ = 0;
   size_t path_bytes_ = 0;
   std::vector<Overlap*> overlaps_;
+  struct PendingSpan {
+    const char* q;
+    const char* t;
+  };
+  std::vector<PendingSpan> pending_;  // parallel to overlaps_; consumed by pack()
+  size_t packed_upto_ = 0;
 };
 
 }  // namespace rga::hip

@@ -7,6 +7,7 @@
 #include <hip/hip_runtime.h>
 
 #include <atomic>
+#include <chrono>
 #include <memory>
 #include <mutex>
 #include <thread>
@@ -60,11 +61,14 @@ class HipPolisher : public Polisher {
     std::mutex queue_mutex;
     uint64_t next_overlap = 0;
     std::atomic<uint64_t> skipped{0};
+    std::atomic<int64_t> t_fill_ns{0}, t_gpu_ns{0};
 
     auto worker = [&](hip::AlignerBatch* batch) {
+      using clk = std::chrono::steady_clock;
       while (true) {
         batch->reset();
         uint32_t pulled = 0;
+        auto t0 = clk::now();
         {
           std::lock_guard<std::mutex> lock(queue_mutex);
           while (next_overlap < overlaps.size()) {
@@ -85,10 +89,13 @@ class HipPolisher : public Polisher {
             }
           }
         }
+        auto t1 = clk::now();
+        t_fill_ns += (t1 - t0).count();
         if (pulled == 0) {
           return;
         }
         skipped += batch->align_and_emit();
+        t_gpu_ns += (clk::now() - t1).count();
       }
     };
 
@@ -102,6 +109,10 @@ class HipPolisher : public Polisher {
     }
     batches.clear();
 
+    fprintf(stderr,
+            "[racon::HipPolisher] align timings: queue %.3f s, pack+gpu+cigar %.3f s "
+            "(sum over %zu batch threads)\n",
+            t_fill_ns.load() / 1e9, t_gpu_ns.load() / 1e9, threads.size());
     if (skipped.load() > 0) {
       fprintf(stderr, "[racon::HipPolisher] %lu overlap(s) aligned on CPU\n",
               static_cast<unsigned long>(skipped.load()));
@@ -138,12 +149,15 @@ class HipPolisher : public Polisher {
     std::mutex queue_mutex;
     uint64_t next_window = 0;
     std::mutex status_mutex;
+    std::atomic<int64_t> t_fill_ns{0}, t_gpu_ns{0};
 
     auto worker = [&](hip::PoaBatch* batch) {
+      using clk = std::chrono::steady_clock;
       std::vector<uint64_t> batch_indices;
       while (true) {
         batch_indices.clear();
         batch->reset();
+        auto t0 = clk::now();
         {
           std::lock_guard<std::mutex> lock(queue_mutex);
           while (next_window < windows_.size()) {
@@ -166,10 +180,13 @@ class HipPolisher : public Polisher {
             }
           }
         }
+        auto t1 = clk::now();
+        t_fill_ns += (t1 - t0).count();
         if (batch_indices.empty()) {
           return;
         }
         auto status = batch->generate(config_.trim);
+        t_gpu_ns += (clk::now() - t1).count();
         {
           std::lock_guard<std::mutex> lock(status_mutex);
           for (size_t i = 0; i < batch_indices.size(); ++i) {
@@ -188,6 +205,10 @@ class HipPolisher : public Polisher {
     for (auto& t : threads) {
       t.join();
     }
+    fprintf(stderr,
+            "[racon::HipPolisher] poa timings: queue %.3f s, pack+gpu+post %.3f s "
+            "(sum over %zu batch threads)\n",
+            t_fill_ns.load() / 1e9, t_gpu_ns.load() / 1e9, threads.size());
 
     // CPU fallback for every window the GPU did not polish
     // (reference cudapolisher.cpp:354-383)

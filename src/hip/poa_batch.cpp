@@ -195,41 +195,54 @@ bool PoaBatch::add_window(const std::shared_ptr<Window>& window, bool* never_fit
     return false;  // arena full; try the next batch round
   }
 
-  // pack
+  // reserve (bookkeeping only; copies happen in pack() outside the queue lock)
   const uint32_t win_idx = static_cast<uint32_t>(windows_.size());
   PoaWindowDesc desc;
   desc.seq_offset = static_cast<uint32_t>(seq_bytes_);
   desc.scratch_idx = win_idx;
+  desc.num_seqs = shipped;
   h_layer_index_[win_idx] = static_cast<uint32_t>(num_layer_ends_);
-
-  uint32_t rel_end = 0;
-  uint32_t packed = 0;
-  for (uint32_t k = 0; k < total_layers && packed < max_depth_ + 1; ++k) {
-    uint32_t i = order[k];
-    auto seq = window->sequence(i);
-    auto qual = window->quality(i);
-    if (k > 0 && seq.second + 1 > limits_.matrix_width) {
-      continue;
-    }
-    std::memcpy(h_seq_ + seq_bytes_, seq.first, seq.second);
-    if (qual.first != nullptr) {
-      for (uint32_t b = 0; b < seq.second; ++b) {
-        h_weight_[seq_bytes_ + b] = static_cast<uint8_t>(qual.first[b]) - 33;
-      }
-    } else {
-      std::memset(h_weight_ + seq_bytes_, 1, seq.second);
-    }
-    seq_bytes_ += seq.second;
-    rel_end += seq.second;
-    h_layer_ends_[num_layer_ends_++] = rel_end;
-    ++packed;
-  }
-  desc.num_seqs = packed;
+  seq_bytes_ += bytes;
+  num_layer_ends_ += shipped;
   h_desc_[win_idx] = desc;
 
   windows_.emplace_back(window);
+  pending_orders_.emplace_back(std::move(order));
   seqs_added_.emplace_back(shipped - 1);  // layers only (effective coverage)
   return true;
+}
+
+void PoaBatch::pack() {
+  for (; packed_upto_ < windows_.size(); ++packed_upto_) {
+    const auto& window = windows_[packed_upto_];
+    const auto& order = pending_orders_[packed_upto_];
+    const PoaWindowDesc& desc = h_desc_[packed_upto_];
+    size_t off = desc.seq_offset;
+    uint32_t ends_at = h_layer_index_[packed_upto_];
+    uint32_t rel_end = 0;
+    uint32_t packed = 0;
+    const uint32_t total_layers = window->num_layers();
+    for (uint32_t k = 0; k < total_layers && packed < desc.num_seqs; ++k) {
+      uint32_t i = order[k];
+      auto seq = window->sequence(i);
+      auto qual = window->quality(i);
+      if (k > 0 && seq.second + 1 > limits_.matrix_width) {
+        continue;
+      }
+      std::memcpy(h_seq_ + off, seq.first, seq.second);
+      if (qual.first != nullptr) {
+        for (uint32_t b = 0; b < seq.second; ++b) {
+          h_weight_[off + b] = static_cast<uint8_t>(qual.first[b]) - 33;
+        }
+      } else {
+        std::memset(h_weight_ + off, 1, seq.second);
+      }
+      off += seq.second;
+      rel_end += seq.second;
+      h_layer_ends_[ends_at++] = rel_end;
+      ++packed;
+    }
+  }
 }
 
 std::vector<bool> PoaBatch::generate(bool trim) {
@@ -238,6 +251,7 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     return polished;
   }
 
+  pack();
   RGA_HIP_CHECK(hipSetDevice(device_));
   auto s = static_cast<hipStream_t>(stream_);
   auto d = [&](const void* dst, const void* src, size_t bytes) {
@@ -304,6 +318,8 @@ std::vector<bool> PoaBatch::generate(bool trim) {
 void PoaBatch::reset() {
   windows_.clear();
   seqs_added_.clear();
+  pending_orders_.clear();
+  packed_upto_ = 0;
   seq_bytes_ = 0;
   num_layer_ends_ = 0;
 }

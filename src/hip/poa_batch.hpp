@@ -24,10 +24,16 @@ class PoaBatch {
   PoaBatch(const PoaBatch&) = delete;
   PoaBatch& operator=(const PoaBatch&) = delete;
 
-  // Packs the window's layers (sorted, CPU-identical order) into the arena.
-  // Returns false when the batch is full (caller retries with a fresh batch)
-  // or the window cannot fit this configuration at all (never_fits set).
+  // Reserves arena space for the window's layers (sorted, CPU-identical
+  // order). Bookkeeping only, callable under the shared queue lock; the
+  // copies happen in pack() / generate(). Returns false when the batch is
+  // full (caller retries with a fresh batch) or the window cannot fit this
+  // configuration at all (never_fits set).
   bool add_window(const std::shared_ptr<Window>& window, bool* never_fits);
+
+  // Copies every reserved window into the pinned staging buffers; called
+  // outside the queue lock (generate() calls it implicitly).
+  void pack();
 
   uint32_t size() const { return static_cast<uint32_t>(windows_.size()); }
   uint32_t capacity() const { return num_slabs_; }
@@ -67,6 +73,8 @@ class PoaBatch {
   size_t num_layer_ends_ = 0;
   std::vector<std::shared_ptr<Window>> windows_;
   std::vector<uint32_t> seqs_added_;  // layers shipped per window (coverage)
+  std::vector<std::vector<uint32_t>> pending_orders_;  // layer order per reserved window
+  size_t packed_upto_ = 0;
 };
 
 }  // namespace rga::hip
