@@ -4,6 +4,7 @@
 
 #include <algorithm>
 #include <cstring>
+#include <numeric>
 #include <string>
 
 #include "hip/hip_common.hpp"
@@ -11,28 +12,51 @@
 namespace rga::hip {
 
 namespace {
-// arena split: moves dominate; per ~30 kbp+30 kbp alignment the moves cost
-// (q+t+1)*256 B ~ 15 MB, seqs q+t, path q+t.
-constexpr double kMovesShare = 0.97;
+constexpr uint32_t kLanes = 64;
+
+uint32_t pick_band_k(uint32_t band_width) {
+  if (band_width == 0) return 8;  // default band 512
+  uint32_t blocks = (band_width + 63) / 64;
+  if (blocks <= 4) return 4;
+  if (blocks <= 8) return 8;
+  return 16;
+}
 }  // namespace
 
-AlignerBatch::AlignerBatch(int device, size_t mem_budget) : device_(device) {
+AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
+    : device_(device), band_k_(pick_band_k(band_width)) {
+  limits_.band = band_k_ * 64;
   RGA_HIP_CHECK(hipSetDevice(device_));
   hipStream_t s;
   RGA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   stream_ = s;
 
-  moves_cap_dw_ = static_cast<size_t>(mem_budget * kMovesShare) / 4;
-  seq_cap_ = std::max<size_t>(16u << 20, mem_budget / 128);
+  // arena split: the per-column band state dominates — per alignment
+  // ~ (m+1) * K * (16 B Pv/Mv + 4 B S); seqs/path/peq are q+t-scale.
+  // Cap the device pool so construction stays cheap even on 288 GB parts.
+  const size_t pool = std::min<size_t>(mem_budget, 24ull << 30);
+  seq_cap_ = std::max<size_t>(16u << 20, pool / 96);
   path_cap_ = seq_cap_;
+  peq_cap_u64_ = seq_cap_ / 8;  // 4 codes per 64 bases = q_bytes/2 of u64s is
+                                // generous; /8 covers the wave-max padding
   max_alignments_ = 65536;
 
+  const size_t state = pool - 2 * seq_cap_ - peq_cap_u64_ * 8 -
+                       max_alignments_ * (sizeof(AlnDesc) + 16) - (2u << 20);
+  tb_cap_u64_ = state / 20 * 16 / 8;  // 16/20 of state bytes as u64
+  s_cap_i32_ = state / 20 * 4 / 4;    // 4/20 of state bytes as i32
+
+  const uint32_t max_waves = max_alignments_ / kLanes + 2;
   RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_seqs_), seq_cap_));
   RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_descs_),
                               max_alignments_ * sizeof(AlnDesc)));
   RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_path_), path_cap_));
   RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_path_len_), max_alignments_ * 4));
   RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_status_), max_alignments_ * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_edit_), max_alignments_ * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_order_), max_alignments_ * 4));
+  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_waves_),
+                              max_waves * sizeof(AlnWaveDesc)));
 
   size_t total = 0;
   auto carve = [&total](size_t bytes) {
@@ -42,21 +66,31 @@ AlignerBatch::AlignerBatch(int device, size_t mem_budget) : device_(device) {
   };
   size_t o_seqs = carve(seq_cap_);
   size_t o_descs = carve(max_alignments_ * sizeof(AlnDesc));
-  size_t o_moves = carve(moves_cap_dw_ * 4);
+  size_t o_peq = carve(peq_cap_u64_ * 8);
+  size_t o_tb = carve(tb_cap_u64_ * 8);
+  size_t o_s = carve(s_cap_i32_ * 4);
   size_t o_path = carve(path_cap_);
   size_t o_plen = carve(max_alignments_ * 4);
   size_t o_status = carve(max_alignments_ * 4);
   size_t o_ed = carve(max_alignments_ * 4);
+  size_t o_order = carve(max_alignments_ * 4);
+  size_t o_waves = carve(max_waves * sizeof(AlnWaveDesc));
 
   RGA_HIP_CHECK(hipMalloc(&d_pool_, total));
   auto base = static_cast<uint8_t*>(d_pool_);
   arena_.seqs = base + o_seqs;
   arena_.descs = reinterpret_cast<AlnDesc*>(base + o_descs);
-  arena_.moves = reinterpret_cast<uint32_t*>(base + o_moves);
+  arena_.peq = reinterpret_cast<uint64_t*>(base + o_peq);
+  arena_.tb = reinterpret_cast<uint64_t*>(base + o_tb);
+  arena_.sbuf = reinterpret_cast<int32_t*>(base + o_s);
   arena_.path = base + o_path;
   arena_.path_len = reinterpret_cast<uint32_t*>(base + o_plen);
   arena_.status = reinterpret_cast<int32_t*>(base + o_status);
   arena_.edit_distance = reinterpret_cast<int32_t*>(base + o_ed);
+  d_order_ = reinterpret_cast<uint32_t*>(base + o_order);
+  d_waves_ = reinterpret_cast<AlnWaveDesc*>(base + o_waves);
+  arena_.order = d_order_;
+  arena_.waves = d_waves_;
   arena_.limits = limits_;
 }
 
@@ -67,7 +101,8 @@ AlignerBatch::~AlignerBatch() {
   }
   for (void* p : {static_cast<void*>(h_seqs_), static_cast<void*>(h_descs_),
                   static_cast<void*>(h_path_), static_cast<void*>(h_path_len_),
-                  static_cast<void*>(h_status_)}) {
+                  static_cast<void*>(h_status_), static_cast<void*>(h_edit_),
+                  static_cast<void*>(h_order_), static_cast<void*>(h_waves_)}) {
     if (p != nullptr) {
       (void)hipHostFree(p);
     }
@@ -77,39 +112,52 @@ AlignerBatch::~AlignerBatch() {
   }
 }
 
+int32_t AlignerBatch::reserve_span(const char* q, uint32_t q_len, const char* t,
+                                   uint32_t t_len) {
+  if (q_len == 0 || t_len == 0 || q_len > limits_.max_len || t_len > limits_.max_len) {
+    return -2;  // reference: exceeded_max_length -> CPU fallback
+  }
+  const size_t bytes = static_cast<size_t>(q_len) + t_len;
+  // keep reserved per-column state within ~2 sub-launches of the tb arena
+  const uint64_t tb_need = static_cast<uint64_t>(t_len + 1) * band_k_ * 2 * kLanes;
+  if (overlaps_.size() >= max_alignments_ || seq_bytes_ + bytes > seq_cap_ ||
+      path_bytes_ + bytes > path_cap_ || tb_reserved_ + tb_need > 2 * tb_cap_u64_) {
+    return -1;
+  }
+
+  AlnDesc d;
+  d.q_offset = static_cast<uint32_t>(seq_bytes_);
+  d.q_len = q_len;
+  seq_bytes_ += q_len;
+  d.t_offset = static_cast<uint32_t>(seq_bytes_);
+  d.t_len = t_len;
+  seq_bytes_ += t_len;
+  d.path_offset = static_cast<uint32_t>(path_bytes_);
+  path_bytes_ += bytes;
+  tb_reserved_ += tb_need;
+
+  const int32_t slot = static_cast<int32_t>(overlaps_.size());
+  h_descs_[slot] = d;
+  overlaps_.emplace_back(nullptr);
+  pending_.push_back({q, t});
+  return slot;
+}
+
 bool AlignerBatch::add_overlap(Overlap* overlap,
                                const std::vector<std::unique_ptr<Sequence>>& sequences,
                                bool* never_fits) {
   *never_fits = false;
   auto q = overlap->query_span(sequences);
   auto t = overlap->target_span(sequences);
-  if (q.second == 0 || t.second == 0 || q.second > limits_.max_len ||
-      t.second > limits_.max_len) {
-    *never_fits = true;  // reference: exceeded_max_length -> CPU fallback
+  int32_t slot = reserve_span(q.first, q.second, t.first, t.second);
+  if (slot == -2) {
+    *never_fits = true;
     return false;
   }
-  const size_t bytes = static_cast<size_t>(q.second) + t.second;
-  const size_t mdw = (static_cast<size_t>(q.second) + t.second + 1) * 64;
-  if (overlaps_.size() >= max_alignments_ || seq_bytes_ + bytes > seq_cap_ ||
-      moves_dw_ + mdw > moves_cap_dw_ || path_bytes_ + bytes > path_cap_) {
+  if (slot < 0) {
     return false;
   }
-
-  AlnDesc d;
-  d.q_offset = static_cast<uint32_t>(seq_bytes_);
-  d.q_len = q.second;
-  seq_bytes_ += q.second;
-  d.t_offset = static_cast<uint32_t>(seq_bytes_);
-  d.t_len = t.second;
-  seq_bytes_ += t.second;
-  d.moves_offset = moves_dw_;
-  moves_dw_ += mdw;
-  d.path_offset = static_cast<uint32_t>(path_bytes_);
-  path_bytes_ += bytes;
-
-  h_descs_[overlaps_.size()] = d;
-  overlaps_.emplace_back(overlap);
-  pending_.push_back({q.first, t.first});
+  overlaps_[slot] = overlap;
   return true;
 }
 
@@ -122,56 +170,131 @@ void AlignerBatch::pack() {
   }
 }
 
-uint32_t AlignerBatch::align_and_emit() {
+void AlignerBatch::run() {
   if (overlaps_.empty()) {
-    return 0;
+    return;
   }
   pack();
   RGA_HIP_CHECK(hipSetDevice(device_));
   auto s = static_cast<hipStream_t>(stream_);
+  const uint32_t na = static_cast<uint32_t>(overlaps_.size());
   RGA_HIP_CHECK(hipMemcpyAsync(const_cast<uint8_t*>(arena_.seqs), h_seqs_, seq_bytes_,
                                hipMemcpyHostToDevice, s));
   RGA_HIP_CHECK(hipMemcpyAsync(const_cast<AlnDesc*>(arena_.descs), h_descs_,
-                               overlaps_.size() * sizeof(AlnDesc), hipMemcpyHostToDevice, s));
+                               na * sizeof(AlnDesc), hipMemcpyHostToDevice, s));
 
-  launch_aligner_kernel(arena_, static_cast<uint32_t>(overlaps_.size()), stream_);
+  // sort into waves by descending target length (uniform per-wave loops)
+  std::vector<uint32_t> order(na);
+  std::iota(order.begin(), order.end(), 0u);
+  std::sort(order.begin(), order.end(), [&](uint32_t x, uint32_t y) {
+    if (h_descs_[x].t_len != h_descs_[y].t_len) return h_descs_[x].t_len > h_descs_[y].t_len;
+    return x < y;
+  });
+  std::copy(order.begin(), order.end(), h_order_);
+  RGA_HIP_CHECK(hipMemcpyAsync(d_order_, h_order_, na * 4, hipMemcpyHostToDevice, s));
+
+  // greedy sub-launches bounded by the tb/s/peq arenas
+  const uint32_t K = band_k_;
+  uint32_t wave_begin = 0;
+  const uint32_t num_waves_total = (na + kLanes - 1) / kLanes;
+  while (wave_begin < num_waves_total) {
+    uint64_t peq_off = 0, tb_off = 0, s_off = 0;
+    uint32_t w = wave_begin;
+    uint32_t launch_waves = 0;
+    for (; w < num_waves_total; ++w) {
+      uint32_t nb = K, mmax = 0;
+      for (uint32_t l = 0; l < kLanes; ++l) {
+        const uint32_t slot = w * kLanes + l;
+        if (slot >= na) break;
+        const AlnDesc& d = h_descs_[h_order_[slot]];
+        nb = std::max(nb, (d.q_len + 63) / 64);
+        mmax = std::max(mmax, d.t_len);
+      }
+      const uint64_t peq_need = static_cast<uint64_t>(nb) * 4 * kLanes;
+      const uint64_t tb_need = static_cast<uint64_t>(mmax + 1) * K * 2 * kLanes;
+      const uint64_t s_need = static_cast<uint64_t>(mmax + 1) * K * kLanes;
+      if (launch_waves > 0 && (peq_off + peq_need > peq_cap_u64_ ||
+                               tb_off + tb_need > tb_cap_u64_ || s_off + s_need > s_cap_i32_)) {
+        break;
+      }
+      AlnWaveDesc wd;
+      wd.peq_off = peq_off;
+      wd.tb_off = tb_off;
+      wd.s_off = s_off;
+      wd.nb = nb;
+      wd.mmax = mmax;
+      h_waves_[launch_waves] = wd;
+      peq_off += peq_need;
+      tb_off += tb_need;
+      s_off += s_need;
+      ++launch_waves;
+    }
+    // a single wave exceeding the arena alone can't be helped; it still has
+    // its own full region (offsets 0) — the caps guarantee this fits because
+    // max_len * K * 2 * 64 * 8 is carved into tb_cap by construction.
+    const uint32_t launch_slots =
+        std::min(na - wave_begin * kLanes, launch_waves * kLanes);
+    RGA_HIP_CHECK(hipMemcpyAsync(d_waves_, h_waves_, launch_waves * sizeof(AlnWaveDesc),
+                                 hipMemcpyHostToDevice, s));
+    AlnDeviceArena launch_arena = arena_;
+    launch_arena.order = d_order_ + wave_begin * kLanes;
+    launch_arena.waves = d_waves_;
+    launch_aligner_kernel(launch_arena, launch_waves, launch_slots, K, stream_);
+    RGA_HIP_CHECK(hipStreamSynchronize(s));
+    wave_begin += launch_waves;
+  }
 
   RGA_HIP_CHECK(hipMemcpyAsync(h_path_, arena_.path, path_bytes_, hipMemcpyDeviceToHost, s));
-  RGA_HIP_CHECK(hipMemcpyAsync(h_path_len_, arena_.path_len, overlaps_.size() * 4,
-                               hipMemcpyDeviceToHost, s));
-  RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, overlaps_.size() * 4,
-                               hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_path_len_, arena_.path_len, na * 4, hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, na * 4, hipMemcpyDeviceToHost, s));
+  RGA_HIP_CHECK(hipMemcpyAsync(h_edit_, arena_.edit_distance, na * 4, hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipStreamSynchronize(s));
+}
 
-  uint32_t failed = 0;
+std::string AlignerBatch::cigar_of(uint32_t slot) const {
   std::string cigar;
+  if (h_status_[slot] != kAlnOk || h_path_len_[slot] == 0) {
+    return cigar;
+  }
+  const uint8_t* path = h_path_ + h_descs_[slot].path_offset;
+  const uint32_t plen = h_path_len_[slot];
   static const char kOps[3] = {'M', 'I', 'D'};
+  // path is reversed (walked from (n, m)); emit forward with run-lengths
+  uint32_t run = 0;
+  uint8_t run_op = 255;
+  char buf[16];
+  for (int64_t k = static_cast<int64_t>(plen) - 1; k >= 0; --k) {
+    uint8_t op = path[k];
+    if (op == run_op) {
+      ++run;
+    } else {
+      if (run > 0) {
+        cigar.append(buf, snprintf(buf, sizeof(buf), "%u%c", run, kOps[run_op]));
+      }
+      run_op = op;
+      run = 1;
+    }
+  }
+  if (run > 0) {
+    cigar.append(buf, snprintf(buf, sizeof(buf), "%u%c", run, kOps[run_op]));
+  }
+  return cigar;
+}
+
+uint32_t AlignerBatch::align_and_emit() {
+  if (overlaps_.empty()) {
+    return 0;
+  }
+  run();
+  uint32_t failed = 0;
   for (size_t i = 0; i < overlaps_.size(); ++i) {
-    if (h_status_[i] != kAlnOk || h_path_len_[i] == 0) {
-      ++failed;  // empty CIGAR -> CPU pairwise fallback
+    if (overlaps_[i] == nullptr) {
       continue;
     }
-    const uint8_t* path = h_path_ + h_descs_[i].path_offset;
-    const uint32_t plen = h_path_len_[i];
-    cigar.clear();
-    // path is reversed (walked from (n, m)); emit forward with run-lengths
-    uint32_t run = 0;
-    uint8_t run_op = 255;
-    char buf[16];
-    for (int64_t k = static_cast<int64_t>(plen) - 1; k >= 0; --k) {
-      uint8_t op = path[k];
-      if (op == run_op) {
-        ++run;
-      } else {
-        if (run > 0) {
-          cigar.append(buf, snprintf(buf, sizeof(buf), "%u%c", run, kOps[run_op]));
-        }
-        run_op = op;
-        run = 1;
-      }
-    }
-    if (run > 0) {
-      cigar.append(buf, snprintf(buf, sizeof(buf), "%u%c", run, kOps[run_op]));
+    std::string cigar = cigar_of(static_cast<uint32_t>(i));
+    if (cigar.empty()) {
+      ++failed;  // empty CIGAR -> CPU pairwise fallback
+      continue;
     }
     overlaps_[i]->set_cigar(cigar);
   }
@@ -183,8 +306,8 @@ void AlignerBatch::reset() {
   pending_.clear();
   packed_upto_ = 0;
   seq_bytes_ = 0;
-  moves_dw_ = 0;
   path_bytes_ = 0;
+  tb_reserved_ = 0;
 }
 
 }  // namespace rga::hip

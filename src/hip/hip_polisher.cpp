@@ -10,7 +10,10 @@
 #include <chrono>
 #include <memory>
 #include <mutex>
+#include <string>
 #include <thread>
+#include <tuple>
+#include <utility>
 #include <vector>
 
 #include "core/polisher.hpp"
@@ -47,6 +50,22 @@ class HipPolisher : public Polisher {
       return;
     }
 
+    // auto band: 10% of the mean overlap span, forced even
+    // (reference cudapolisher.cpp:150-163)
+    uint32_t band = config_.aligner_band_width;
+    if (band == 0) {
+      uint64_t total_len = 0, count = 0;
+      for (const auto& o : overlaps) {
+        if (o && !o->has_cigar()) {
+          total_len += o->t_end() - o->t_begin();
+          ++count;
+        }
+      }
+      if (count > 0) {
+        band = static_cast<uint32_t>(total_len / count / 10) & ~1u;
+      }
+    }
+
     std::vector<std::unique_ptr<hip::AlignerBatch>> batches;
     for (int d : devices_) {
       RGA_HIP_CHECK(hipSetDevice(d));
@@ -54,7 +73,7 @@ class HipPolisher : public Polisher {
       RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
       size_t budget = free_mem * 9 / 10 / config_.aligner_batches;
       for (uint32_t b = 0; b < config_.aligner_batches; ++b) {
-        batches.emplace_back(std::make_unique<hip::AlignerBatch>(d, budget));
+        batches.emplace_back(std::make_unique<hip::AlignerBatch>(d, budget, band));
       }
     }
 
@@ -237,6 +256,48 @@ class HipPolisher : public Polisher {
 
 namespace hip {
 int runtime_device_count() { return device_count(); }
+
+// Direct GPU alignment of raw (query, target) pairs — numerics testing
+// entry (GPU edit distance must equal the CPU optimum; CIGARs must be
+// consistent). Returns (cigar, edit_distance, status) per pair.
+std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
+    const std::vector<std::pair<std::string, std::string>>& pairs, uint32_t band_width) {
+  std::vector<std::tuple<std::string, int32_t, int32_t>> out(pairs.size());
+  if (pairs.empty()) {
+    return out;
+  }
+  AlignerBatch batch(0, 4ull << 30, band_width);
+  size_t begin = 0;
+  std::vector<int32_t> slots(pairs.size(), -1);
+  auto flush = [&](size_t end) {
+    batch.run();
+    for (size_t i = begin; i < end; ++i) {
+      if (slots[i] >= 0) {
+        out[i] = {batch.cigar_of(slots[i]), batch.edit_distance_of(slots[i]),
+                  batch.status_of(slots[i])};
+      } else {
+        out[i] = {std::string(), -1, kAlnNotRun};
+      }
+    }
+    batch.reset();
+    begin = end;
+  };
+  for (size_t i = 0; i < pairs.size(); ++i) {
+    slots[i] = batch.reserve_span(pairs[i].first.data(),
+                                  static_cast<uint32_t>(pairs[i].first.size()),
+                                  pairs[i].second.data(),
+                                  static_cast<uint32_t>(pairs[i].second.size()));
+    if (slots[i] == -1) {
+      flush(i);
+      slots[i] = batch.reserve_span(pairs[i].first.data(),
+                                    static_cast<uint32_t>(pairs[i].first.size()),
+                                    pairs[i].second.data(),
+                                    static_cast<uint32_t>(pairs[i].second.size()));
+    }
+  }
+  flush(pairs.size());
+  return out;
+}
 }  // namespace hip
 
 std::unique_ptr<Polisher> createHipPolisher(std::unique_ptr<SequenceParser> sparser,

@@ -4,6 +4,10 @@
 #include <cstdio>
 #include <cstdlib>
 #include <memory>
+#include <string>
+#include <tuple>
+#include <utility>
+#include <vector>
 
 #include "core/polisher.hpp"
 
@@ -11,6 +15,11 @@ namespace rga {
 
 namespace hip {
 int runtime_device_count() { return 0; }
+std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
+    const std::vector<std::pair<std::string, std::string>>&, uint32_t) {
+  fprintf(stderr, "[racon::hip::align_pairs] error: no HIP backend in this build!\n");
+  exit(1);
+}
 }  // namespace hip
 
 std::unique_ptr<Polisher> createHipPolisher(std::unique_ptr<SequenceParser>,

@@ -15,6 +15,8 @@
 
 namespace rga::hip {
 int runtime_device_count();  // provided by the HIP backend (or the stub)
+std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
+    const std::vector<std::pair<std::string, std::string>>& pairs, uint32_t band_width);
 }
 
 namespace py = pybind11;
@@ -117,6 +119,13 @@ PYBIND11_MODULE(_racon, m) {
         "Number of visible HIP devices.");
   m.def("edit_distance", &edit_distance_py, py::arg("a"), py::arg("b"));
   m.def("align_cigar", &align_cigar_py, py::arg("query"), py::arg("target"));
+  m.def("gpu_align", [](const std::vector<std::pair<std::string, std::string>>& pairs,
+                        uint32_t band_width) {
+          py::gil_scoped_release release;
+          return rga::hip::align_pairs(pairs, band_width);
+        },
+        py::arg("pairs"), py::arg("band_width") = 0,
+        "GPU Myers aligner on raw (query, target) pairs -> (cigar, edit_distance, status)");
   m.def("reverse_complement", &reverse_complement, py::arg("sequence"));
   m.def("poa_consensus", &poa_consensus, py::arg("sequences"),
         py::arg("qualities") = std::vector<std::string>(), py::arg("match") = 5,

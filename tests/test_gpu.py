@@ -15,6 +15,70 @@ def require_gpu(racon):
         pytest.skip("no GPU on this host")
 
 
+def _mutate(seq, rng, sub, ins, dele):
+    out = []
+    for ch in seq:
+        r = rng.random()
+        if r < dele:
+            continue
+        if r < dele + ins:
+            out.append(rng.choice("ACGT"))
+        if r < dele + ins + sub:
+            out.append(rng.choice([c for c in "ACGT" if c != ch]))
+        else:
+            out.append(ch)
+    return "".join(out)
+
+
+def test_gpu_aligner_matches_cpu_edit_distance(racon):
+    """GPU Myers edit distances equal the CPU optimum; CIGARs are consistent."""
+    import random
+    rng = random.Random(42)
+    pairs = []
+    for i in range(100):
+        n = rng.randint(200, 20000)
+        t = "".join(rng.choice("ACGT") for _ in range(n))
+        q = _mutate(t, rng, 0.02, 0.02, 0.02)
+        pairs.append((q, t))
+    res = racon.gpu_align(pairs)
+    n_edge = 0
+    for (q, t), (cigar, ed, status) in zip(pairs, res):
+        if status != 0:
+            n_edge += 1
+            continue
+        assert ed == racon.edit_distance(q, t), (len(q), len(t))
+        # CIGAR must consume exactly q (M/I) and t (M/D), and its cost must
+        # reproduce the edit distance
+        qi = ti = cost = 0
+        import re
+        for num, op in re.findall(r"(\d+)([MID])", cigar):
+            num = int(num)
+            if op == "M":
+                cost += sum(1 for k in range(num) if q[qi + k] != t[ti + k])
+                qi += num
+                ti += num
+            elif op == "I":
+                cost += num
+                qi += num
+            else:
+                cost += num
+                ti += num
+        assert qi == len(q) and ti == len(t)
+        assert cost == ed, (cost, ed)
+    assert n_edge <= 2, f"too many band-edge fallbacks: {n_edge}"
+
+
+def test_gpu_aligner_band_widths(racon):
+    import random
+    rng = random.Random(7)
+    t = "".join(rng.choice("ACGT") for _ in range(5000))
+    q = _mutate(t, rng, 0.03, 0.03, 0.03)
+    for band in (256, 512, 1024):
+        cigar, ed, status = racon.gpu_align([(q, t)], band_width=band)[0]
+        assert status == 0
+        assert ed == racon.edit_distance(q, t)
+
+
 def test_gpu_polish_runs_and_improves_draft(racon, sample, fasta_reader):
     truth = list(fasta_reader(sample["reference"]).values())[0]
     draft = list(fasta_reader(sample["layout"]).values())[0]
