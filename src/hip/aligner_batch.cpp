@@ -21,6 +21,13 @@ uint32_t pick_band_k(uint32_t band_width) {
   if (blocks <= 8) return 8;
   return 16;
 }
+
+// The reference's auto heuristic (10% of mean overlap length,
+// cudapolisher.cpp:159-163) is tuned for its approximate banded DP; the
+// Myers band here is exact within the band and indel drift on long reads is
+// a short random walk, so clamp the auto choice to [256, 512]. Explicit
+// --cudaaligner-band-width values above 512 still select K=16.
+
 }  // namespace
 
 AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
@@ -118,10 +125,11 @@ int32_t AlignerBatch::reserve_span(const char* q, uint32_t q_len, const char* t,
     return -2;  // reference: exceeded_max_length -> CPU fallback
   }
   const size_t bytes = static_cast<size_t>(q_len) + t_len;
-  // keep reserved per-column state within ~2 sub-launches of the tb arena
-  const uint64_t tb_need = static_cast<uint64_t>(t_len + 1) * band_k_ * 2 * kLanes;
+  // per-lane share of the wave's tb region (64 alignments share one wave);
+  // cap at ~2 sub-launches' worth, with headroom for wave-max padding
+  const uint64_t tb_need = static_cast<uint64_t>(t_len + 1) * band_k_ * 2;
   if (overlaps_.size() >= max_alignments_ || seq_bytes_ + bytes > seq_cap_ ||
-      path_bytes_ + bytes > path_cap_ || tb_reserved_ + tb_need > 2 * tb_cap_u64_) {
+      path_bytes_ + bytes > path_cap_ || tb_reserved_ + tb_need > 2 * (tb_cap_u64_ / 64)) {
     return -1;
   }
 

@@ -62,7 +62,11 @@ class HipPolisher : public Polisher {
         }
       }
       if (count > 0) {
+        // 10% of mean span like the reference, clamped: the Myers band is
+        // exact inside the band and the CPU fallback catches escapes, so a
+        // tight band wins (see pick_band_k in aligner_batch.cpp)
         band = static_cast<uint32_t>(total_len / count / 10) & ~1u;
+        band = band < 256 ? 256 : (band > 512 ? 512 : band);
       }
     }
 

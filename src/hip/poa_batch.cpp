@@ -23,6 +23,7 @@ size_t slab_bytes(const PoaLimits& L) {
   b += n * 12;                      // hb_score(8) + hb_pred(4)
   b += (2 * L.matrix_width + n) * 8;  // aln_nodes + aln_seq
   b += (n + 1) * L.matrix_width * 2;  // matrix
+  b += (n + 1) * L.matrix_width;      // moves
   return b;
 }
 
@@ -104,6 +105,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   size_t o_aln_n = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
   size_t o_aln_s = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
   size_t o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
+  size_t o_moves = carve(num_slabs_ * (n + 1) * L.matrix_width);
   size_t o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
   size_t o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
   size_t o_clen = carve(num_slabs_ * 4);
@@ -133,6 +135,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.aln_nodes = reinterpret_cast<int32_t*>(base + o_aln_n);
   arena_.aln_seq = reinterpret_cast<int32_t*>(base + o_aln_s);
   arena_.matrix = reinterpret_cast<int16_t*>(base + o_matrix);
+  arena_.moves = base + o_moves;
   arena_.consensus = base + o_cons;
   arena_.coverage = reinterpret_cast<uint16_t*>(base + o_cov);
   arena_.consensus_len = reinterpret_cast<uint32_t*>(base + o_clen);

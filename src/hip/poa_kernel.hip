@@ -1,14 +1,19 @@
-// HIP/CDNA4 POA mega-kernel: one 64-lane wavefront per window.
+// HIP/CDNA4 POA mega-kernel v2: one 64-lane wavefront per window, LDS-centric.
 //
-// Per window it loops over the layers: banded/full Needleman-Wunsch of the
-// layer against the current POA graph (rows = graph nodes in topological
-// order, columns = layer bases; the horizontal gap pass is a wave-wide
-// max-scan with slope), traceback, threading the alignment into the graph,
-// Kahn topological re-sort, and finally heaviest-bundle consensus with
-// per-base coverage. DP rows are streamed to HBM as int16 (coalesced 64-lane
-// chunks); the serial graph phases run on lane 0 and are hidden by the 16-32
-// co-resident windows per CU. No __syncthreads in the hot loop: a single
-// wavefront is synchronous by construction.
+// Per window it loops over the layers: Needleman-Wunsch of the layer against
+// the current POA graph (rows = graph nodes in topological order, columns =
+// layer bases; the horizontal gap pass is a wave-wide max-scan with slope),
+// move-byte traceback, threading the alignment into the graph, Kahn
+// topological re-sort, and finally heaviest-bundle consensus with per-base
+// coverage.
+//
+// v2 vs v1 (v1 streamed every DP row and all graph arrays through HBM and
+// measured 85% of wave-cycles parked on memory waits): the hot graph arrays
+// (letters / rank / sorted / in- & out-counts / first edges) and a 4-row DP
+// ring live in LDS (~31 KB/block, 5 blocks/CU); predecessor rows are LDS
+// hits unless the rank distance exceeds the ring (rare bubbles -> global
+// matrix). The DP records a move byte (type + in-edge index) per cell so the
+// serial traceback is one byte load per step instead of score re-derivation.
 //
 // Semantics mirror the CPU engine (src/align/poa.cpp) so GPU results are
 // deterministic and window-content-only (reference racon-gpu pins separate
@@ -23,6 +28,15 @@ namespace {
 
 constexpr int kLanes = 64;
 constexpr int32_t kNegInf = -(1 << 28);
+constexpr uint32_t kMaxN = 2048;   // LDS mirrors; PoaLimits.max_nodes must fit
+constexpr uint32_t kMaxW = 1024;   // LDS row width; matrix_width must fit
+constexpr uint32_t kRing = 4;      // DP rows kept in LDS
+
+// move byte encoding
+constexpr uint8_t kMvDiag = 0;
+constexpr uint8_t kMvUp = 1;
+constexpr uint8_t kMvLeft = 2;
+constexpr uint8_t kMvInvalid = 3;
 
 __device__ inline int32_t wave_scan_max(int32_t v, int lane) {
   // inclusive max-scan over the 64-lane wavefront
@@ -35,25 +49,39 @@ __device__ inline int32_t wave_scan_max(int32_t v, int lane) {
   return v;
 }
 
+// LDS mirrors of the hot graph arrays + DP row ring. `work` (Kahn scratch)
+// aliases the DP ring: the ring's contents are only read within one layer's
+// DP (every predecessor row is recomputed before use), so the sort may
+// clobber it freely between layers.
+struct Shared {
+  union {
+    int16_t ring[kRing][kMaxW];  // DP rows r-3..r (slot = row % kRing)
+    uint16_t work[kMaxN];        // Kahn in-degree scratch
+  } u;
+  uint16_t rank[kMaxN];
+  uint16_t sorted[kMaxN];
+  uint16_t first_in[kMaxN];
+  uint16_t first_out[kMaxN];
+  uint8_t letters[kMaxN];
+  uint8_t in_cnt[kMaxN];
+  uint8_t out_cnt[kMaxN];
+  uint8_t seq[kMaxW];
+};
+
 struct WindowCtx {
-  // slab pointers for this window
-  uint8_t* letters;
-  uint8_t* in_cnt;
-  uint8_t* out_cnt;
-  uint8_t* ring_cnt;
+  // global slab pointers for this window (cold arrays)
   uint16_t* in_edges;
   int32_t* in_weights;
   uint16_t* out_edges;
+  uint8_t* ring_cnt;
   uint16_t* ring;
   uint16_t* nseq;
-  uint16_t* sorted;
-  uint16_t* rank;
-  uint16_t* work;
   int64_t* hb_score;
   int32_t* hb_pred;
   int32_t* aln_nodes;
   int32_t* aln_seq;
   int16_t* matrix;
+  uint8_t* moves;
 
   const uint8_t* seq_base;
   const uint8_t* weight_base;
@@ -73,14 +101,23 @@ struct WindowCtx {
 
 // ---------- serial (lane 0) graph helpers ----------
 
-__device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t w) {
-  uint32_t n_out = c.out_cnt[a];
+__device__ inline uint16_t in_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
+                                      uint32_t e) {
+  return (e == 0) ? s.first_in[node] : c.in_edges[node * c.ME + e];
+}
+
+__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
+                                       uint32_t e) {
+  return (e == 0) ? s.first_out[node] : c.out_edges[node * c.ME + e];
+}
+
+__device__ bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
+  uint32_t n_out = s.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
-    if (c.out_edges[a * c.ME + e] == b) {
-      // find matching in-edge slot on b to bump the weight
-      uint32_t n_in = c.in_cnt[b];
+    if (out_edge_of(c, s, a, e) == b) {
+      uint32_t n_in = s.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
-        if (c.in_edges[b * c.ME + f] == a) {
+        if (in_edge_of(c, s, b, f) == a) {
           c.in_weights[b * c.ME + f] += w;
           return true;
         }
@@ -88,28 +125,34 @@ __device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t 
       return true;  // unreachable for a consistent graph
     }
   }
-  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
+  if (n_out >= c.ME || s.in_cnt[b] >= c.ME) {
     c.status = kPoaEdgeOverflow;
     return false;
   }
   c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
-  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
-  uint32_t n_in = c.in_cnt[b];
+  if (n_out == 0) {
+    s.first_out[a] = static_cast<uint16_t>(b);
+  }
+  s.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = s.in_cnt[b];
   c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
+  if (n_in == 0) {
+    s.first_in[b] = static_cast<uint16_t>(a);
+  }
   c.in_weights[b * c.ME + n_in] = w;
-  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  s.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
-  if (c.num_nodes >= c.MN) {
+__device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
+  if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
     c.status = kPoaNodeOverflow;
     return -1;
   }
   uint32_t id = c.num_nodes++;
-  c.letters[id] = letter;
-  c.in_cnt[id] = 0;
-  c.out_cnt[id] = 0;
+  s.letters[id] = letter;
+  s.in_cnt[id] = 0;
+  s.out_cnt[id] = 0;
   c.ring_cnt[id] = 0;
   c.nseq[id] = 0;
   return static_cast<int32_t>(id);
@@ -117,7 +160,7 @@ __device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
 
 // Threads the traceback path (stored reversed in aln_*) into the graph.
 // Mirrors Graph::add_alignment (src/align/poa.cpp).
-__device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t* wts,
+__device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, const uint8_t* wts,
                                 uint32_t len, int32_t aln_len) {
   // first/last aligned sequence positions
   int32_t first_pos = -1, last_pos = -1;
@@ -135,7 +178,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
   int32_t last_counted = -1;
 
   auto link = [&](int32_t a, int32_t b, int32_t w) {
-    if (!add_edge_d(c, a, b, w)) {
+    if (!add_edge_d(c, s, a, b, w)) {
       return;
     }
     if (a != last_counted) {
@@ -152,7 +195,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
 
   // head chain: seq[0 .. first_pos)
   for (int32_t p = 0; p < first_pos; ++p) {
-    int32_t id = add_node_d(c, seq[p]);
+    int32_t id = add_node_d(c, s, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -171,22 +214,22 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
     int32_t node = c.aln_nodes[k];
     int32_t new_id;
     if (node == -1) {
-      new_id = add_node_d(c, letter);
+      new_id = add_node_d(c, s, letter);
       if (new_id < 0) return;
-    } else if (c.letters[node] == letter) {
+    } else if (s.letters[node] == letter) {
       new_id = node;
     } else {
       new_id = -1;
       uint32_t nr = c.ring_cnt[node];
       for (uint32_t r = 0; r < nr; ++r) {
         uint16_t aid = c.ring[node * c.MR + r];
-        if (c.letters[aid] == letter) {
+        if (s.letters[aid] == letter) {
           new_id = aid;
           break;
         }
       }
       if (new_id == -1) {
-        new_id = add_node_d(c, letter);
+        new_id = add_node_d(c, s, letter);
         if (new_id < 0) return;
         // join the ring: new node linked with node and all its partners
         if (nr >= c.MR) {
@@ -220,7 +263,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
 
   // tail chain: seq[last_pos+1 .. len)
   for (int32_t p = (last_pos == -1 ? len : last_pos + 1); p < static_cast<int32_t>(len); ++p) {
-    int32_t id = add_node_d(c, seq[p]);
+    int32_t id = add_node_d(c, s, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -232,48 +275,49 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
   ++c.seqs_in_graph;
 }
 
-// Kahn topological sort (FIFO, deterministic). Rebuilds sorted/rank.
-__device__ void topo_sort_d(WindowCtx& c) {
+// Kahn topological sort (FIFO, deterministic) over the LDS mirrors.
+__device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
-    c.work[i] = c.in_cnt[i];
+    s.u.work[i] = s.in_cnt[i];
   }
   uint32_t qhead = 0, qtail = 0;
   for (uint32_t i = 0; i < n; ++i) {
-    if (c.work[i] == 0) {
-      c.sorted[qtail++] = static_cast<uint16_t>(i);
+    if (s.u.work[i] == 0) {
+      s.sorted[qtail++] = static_cast<uint16_t>(i);
     }
   }
   while (qhead < qtail) {
-    uint16_t u = c.sorted[qhead++];
-    uint32_t nout = c.out_cnt[u];
+    uint16_t u = s.sorted[qhead++];
+    uint32_t nout = s.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t v = c.out_edges[u * c.ME + e];
-      if (--c.work[v] == 0) {
-        c.sorted[qtail++] = v;
+      uint16_t v = out_edge_of(c, s, u, e);
+      if (--s.u.work[v] == 0) {
+        s.sorted[qtail++] = v;
       }
     }
   }
   for (uint32_t r = 0; r < qtail; ++r) {
-    c.rank[c.sorted[r]] = static_cast<uint16_t>(r);
+    s.rank[s.sorted[r]] = static_cast<uint16_t>(r);
   }
 }
 
 // Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
 // Returns consensus length written into out/cov (forward order), or -1.
-__device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32_t max_out) {
+__device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* cov,
+                               uint32_t max_out) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
     c.hb_score[i] = -1;
     c.hb_pred[i] = -1;
   }
 
-  uint32_t max_id = c.sorted[0];
+  uint32_t max_id = s.sorted[0];
   for (uint32_t r = 0; r < n; ++r) {
-    uint16_t nid = c.sorted[r];
-    uint32_t nin = c.in_cnt[nid];
+    uint16_t nid = s.sorted[r];
+    uint32_t nin = s.in_cnt[nid];
     for (uint32_t e = 0; e < nin; ++e) {
-      uint16_t p = c.in_edges[nid * c.ME + e];
+      uint16_t p = in_edge_of(c, s, nid, e);
       int64_t w = c.in_weights[nid * c.ME + e];
       if (c.hb_score[nid] < w ||
           (c.hb_score[nid] == w && c.hb_pred[nid] != -1 &&
@@ -295,20 +339,20 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
   // otherwise spin forever — those windows fail over to the CPU instead)
   uint32_t guard = 0;
   uint32_t prev_rank = 0;
-  while (c.out_cnt[max_id] != 0) {
-    if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
+  while (s.out_cnt[max_id] != 0) {
+    if (++guard > n || (guard > 1 && s.rank[max_id] <= prev_rank)) {
       c.status = kPoaConsensusOverflow;
       return -1;
     }
-    prev_rank = c.rank[max_id];
-    uint32_t rank0 = c.rank[max_id];
+    prev_rank = s.rank[max_id];
+    uint32_t rank0 = s.rank[max_id];
     // invalidate alternative branches
-    uint32_t nout = c.out_cnt[max_id];
+    uint32_t nout = s.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t endn = c.out_edges[max_id * c.ME + e];
-      uint32_t nin = c.in_cnt[endn];
+      uint16_t endn = out_edge_of(c, s, max_id, e);
+      uint32_t nin = s.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
-        uint16_t o = c.in_edges[endn * c.ME + f];
+        uint16_t o = in_edge_of(c, s, endn, f);
         if (o != max_id) {
           c.hb_score[o] = -1;
         }
@@ -317,12 +361,12 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
     int64_t best = 0;
     uint32_t best_id = 0;
     for (uint32_t r = rank0 + 1; r < n; ++r) {
-      uint16_t nid = c.sorted[r];
+      uint16_t nid = s.sorted[r];
       c.hb_score[nid] = -1;
       c.hb_pred[nid] = -1;
-      uint32_t nin = c.in_cnt[nid];
+      uint32_t nin = s.in_cnt[nid];
       for (uint32_t e = 0; e < nin; ++e) {
-        uint16_t p = c.in_edges[nid * c.ME + e];
+        uint16_t p = in_edge_of(c, s, nid, e);
         if (c.hb_score[p] == -1) {
           continue;
         }
@@ -359,7 +403,7 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
   idw = static_cast<int32_t>(max_id);
   for (int32_t k = path_len - 1; k >= 0; --k) {
     uint32_t node = static_cast<uint32_t>(idw);
-    out[k] = c.letters[node];
+    out[k] = s.letters[node];
     uint32_t covv = c.nseq[node];
     uint32_t nr = c.ring_cnt[node];
     for (uint32_t r = 0; r < nr; ++r) {
@@ -373,7 +417,7 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
 
 // ---------- the mega-kernel ----------
 
-__launch_bounds__(kLanes, 8)
+__launch_bounds__(kLanes)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const uint32_t win = blockIdx.x;
   if (win >= num_windows) {
@@ -384,26 +428,21 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const PoaLimits L = a.limits;
   const uint32_t slab = desc.scratch_idx;
 
-  __shared__ uint8_t lds_seq[1024];
+  __shared__ Shared s;
 
   WindowCtx c;
-  c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
-  c.in_cnt = a.in_cnt + static_cast<size_t>(slab) * L.max_nodes;
-  c.out_cnt = a.out_cnt + static_cast<size_t>(slab) * L.max_nodes;
-  c.ring_cnt = a.ring_cnt + static_cast<size_t>(slab) * L.max_nodes;
   c.in_edges = a.in_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
   c.in_weights = a.in_weights + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
   c.out_edges = a.out_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
+  c.ring_cnt = a.ring_cnt + static_cast<size_t>(slab) * L.max_nodes;
   c.ring = a.ring + static_cast<size_t>(slab) * L.max_nodes * L.max_ring;
   c.nseq = a.nseq + static_cast<size_t>(slab) * L.max_nodes;
-  c.sorted = a.sorted + static_cast<size_t>(slab) * L.max_nodes;
-  c.rank = a.rank + static_cast<size_t>(slab) * L.max_nodes;
-  c.work = a.work + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_score = a.hb_score + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_pred = a.hb_pred + static_cast<size_t>(slab) * L.max_nodes;
   c.aln_nodes = a.aln_nodes + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
   c.aln_seq = a.aln_seq + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
   c.matrix = a.matrix + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
+  c.moves = a.moves + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
 
   c.seq_base = a.seq_data + desc.seq_offset;
   c.weight_base = a.weight_data + desc.seq_offset;
@@ -412,39 +451,41 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.ME = L.max_edges;
   c.MR = L.max_ring;
   c.MW = L.matrix_width;
-  c.MN = L.max_nodes;
+  c.MN = min(L.max_nodes, kMaxN);
   c.m = a.match;
   c.x = a.mismatch;
   c.g = a.gap;
-  c.status = kPoaOk;
+  c.status = (L.matrix_width <= kMaxW) ? kPoaOk : kPoaNodeOverflow;
 
   // ---- init graph from the backbone (layer 0), lane-parallel ----
   const uint32_t bb_len = c.ends[0];
   const uint8_t* bb_seq = c.seq_base;
   const uint8_t* bb_wts = c.weight_base;
   for (uint32_t i = lane; i < bb_len; i += kLanes) {
-    c.letters[i] = bb_seq[i];
+    s.letters[i] = bb_seq[i];
     c.ring_cnt[i] = 0;
     c.nseq[i] = bb_len >= 2 ? 1 : 0;
-    c.sorted[i] = static_cast<uint16_t>(i);
-    c.rank[i] = static_cast<uint16_t>(i);
+    s.sorted[i] = static_cast<uint16_t>(i);
+    s.rank[i] = static_cast<uint16_t>(i);
     if (i == 0) {
-      c.in_cnt[i] = 0;
+      s.in_cnt[i] = 0;
     } else {
-      c.in_cnt[i] = 1;
+      s.in_cnt[i] = 1;
+      s.first_in[i] = static_cast<uint16_t>(i - 1);
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
     if (i + 1 < bb_len) {
-      c.out_cnt[i] = 1;
+      s.out_cnt[i] = 1;
+      s.first_out[i] = static_cast<uint16_t>(i + 1);
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
-      c.out_cnt[i] = 0;
+      s.out_cnt[i] = 0;
     }
   }
   c.num_nodes = bb_len;
   c.seqs_in_graph = 1;
-  __threadfence_block();  // backbone graph writes -> visible to all lanes
+  __syncthreads();  // LDS graph writes -> visible to all lanes
 
   // ---- per-layer loop ----
   for (uint32_t layer = 1; layer < c.num_seqs && c.status == kPoaOk; ++layer) {
@@ -457,43 +498,51 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     }
 
     for (uint32_t j = lane; j < len; j += kLanes) {
-      lds_seq[j] = seq[j];
+      s.seq[j] = seq[j];
     }
+    __syncthreads();
 
     const uint32_t n = c.num_nodes;
     const uint32_t width = len + 1;
     const uint32_t chunks = (len + kLanes - 1) / kLanes;
 
-    // row 0: all-gap prefix of the layer
-    for (uint32_t j = lane; j < width; j += kLanes) {
-      c.matrix[j] = static_cast<int16_t>(static_cast<int32_t>(j) * c.g);
-    }
-    __threadfence_block();
-
     int32_t best_score = kNegInf;
     uint32_t best_row = 0;
 
+    // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
     for (uint32_t r = 0; r < n; ++r) {
-      const uint16_t node = c.sorted[r];
-      const uint8_t letter = c.letters[node];
-      const uint32_t nin = c.in_cnt[node];
+      const uint16_t node = s.sorted[r];
+      const uint8_t letter = s.letters[node];
+      const uint32_t nin = s.in_cnt[node];
       int16_t* Hrow = c.matrix + static_cast<size_t>(r + 1) * c.MW;
+      uint8_t* Mrow = c.moves + static_cast<size_t>(r + 1) * c.MW;
+      int16_t* ring_row = s.u.ring[(r + 1) % kRing];
 
       // first column (j = 0): max over preds of Hp[0] + gap
       int32_t h0;
       {
         int32_t best0 = kNegInf;
+        uint32_t e0 = 0;
         if (nin == 0) {
           best0 = 0;
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
-            uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
-            best0 = max(best0, static_cast<int32_t>(c.matrix[static_cast<size_t>(p) * c.MW]));
+            const uint32_t p = s.rank[in_edge_of(c, s, node, e)] + 1;
+            const int32_t hp0 = (r + 1 - p < kRing)
+                                    ? s.u.ring[p % kRing][0]
+                                    : c.matrix[static_cast<size_t>(p) * c.MW];
+            if (hp0 > best0) {
+              best0 = hp0;
+              e0 = e;
+            }
           }
         }
         h0 = best0 + c.g;
         if (lane == 0) {
           Hrow[0] = static_cast<int16_t>(h0);
+          ring_row[0] = static_cast<int16_t>(h0);
+          // column 0 is always a vertical chain through the argmax edge
+          Mrow[0] = static_cast<uint8_t>(kMvUp | (e0 << 2));
         }
       }
 
@@ -504,17 +553,49 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       for (uint32_t k = 0; k < chunks; ++k) {
         const uint32_t j = 1 + k * kLanes + lane;  // column this lane owns
         int32_t v = kNegInf;
+        uint32_t ve = 0;
+        uint8_t vtype = kMvInvalid;
         if (j < width) {
-          const int32_t sub = (lds_seq[j - 1] == letter) ? c.m : c.x;
+          const int32_t sub = (s.seq[j - 1] == letter) ? c.m : c.x;
           if (nin == 0) {
-            const int16_t* Hp = c.matrix;  // row 0
-            v = max(static_cast<int32_t>(Hp[j - 1]) + sub, static_cast<int32_t>(Hp[j]) + c.g);
+            // pred is the arithmetic row 0
+            const int32_t diag = static_cast<int32_t>(j - 1) * c.g + sub;
+            const int32_t up = static_cast<int32_t>(j) * c.g + c.g;
+            v = max(diag, up);
+            vtype = (diag >= up) ? kMvDiag : kMvUp;
           } else {
+            int32_t best_diag = kNegInf, best_up = kNegInf;
+            uint32_t e_diag = 0, e_up = 0;
             for (uint32_t e = 0; e < nin; ++e) {
-              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
-              const int16_t* Hp = c.matrix + static_cast<size_t>(p) * c.MW;
-              v = max(v, max(static_cast<int32_t>(Hp[j - 1]) + sub,
-                             static_cast<int32_t>(Hp[j]) + c.g));
+              const uint32_t p = s.rank[in_edge_of(c, s, node, e)] + 1;
+              int32_t hpjm1, hpj;
+              if (r + 1 - p < kRing) {
+                const int16_t* Rp = s.u.ring[p % kRing];
+                hpjm1 = Rp[j - 1];
+                hpj = Rp[j];
+              } else {
+                const int16_t* Hp = c.matrix + static_cast<size_t>(p) * c.MW;
+                hpjm1 = Hp[j - 1];
+                hpj = Hp[j];
+              }
+              if (hpjm1 + sub > best_diag) {
+                best_diag = hpjm1 + sub;
+                e_diag = e;
+              }
+              if (hpj + c.g > best_up) {
+                best_up = hpj + c.g;
+                e_up = e;
+              }
+            }
+            // traceback priority: diagonal first, then vertical
+            if (best_diag >= best_up) {
+              v = best_diag;
+              vtype = kMvDiag;
+              ve = e_diag;
+            } else {
+              v = best_up;
+              vtype = kMvUp;
+              ve = e_up;
             }
           }
         }
@@ -525,6 +606,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         const int32_t h = u + static_cast<int32_t>(j) * c.g;
         if (j < width) {
           Hrow[j] = static_cast<int16_t>(h);
+          ring_row[j] = static_cast<int16_t>(h);
+          // left move only when the scan strictly beat this cell's v
+          Mrow[j] = (h == v) ? static_cast<uint8_t>(vtype | (ve << 2)) : kMvLeft;
           if (j == len) {
             last_col_val = h;
           }
@@ -532,12 +616,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         carry_u = __shfl(u, kLanes - 1, kLanes);
       }
 
-      // this row's stores must be visible to every lane before the next row
-      // (cross-lane read-after-write through global memory within one wave)
-      __threadfence_block();
+      // ring/global row writes must be visible to every lane before the next
+      // row (cross-lane read-after-write within one wave)
+      __syncthreads();
 
       // end-node max (strict >, first in topological order wins)
-      if (c.out_cnt[node] == 0) {
+      if (s.out_cnt[node] == 0) {
         const int src_lane = static_cast<int>((len - 1) % kLanes);
         const int32_t lc = __shfl(last_col_val, src_lane, kLanes);
         if (lc > best_score) {
@@ -549,79 +633,58 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
 
     // ---- serial phases on lane 0 ----
     if (lane == 0) {
-      // traceback (priority: diagonal preds in order, vertical, horizontal)
+      // move-byte traceback
       int32_t aln_len = 0;
       uint32_t i = best_row, j = len;
       while (!(i == 0 && j == 0)) {
-        const int32_t H_ij = c.matrix[static_cast<size_t>(i) * c.MW + j];
         uint32_t prev_i = i, prev_j = j;
-        bool found = false;
-        if (i != 0 && j != 0) {
-          const uint16_t node = c.sorted[i - 1];
-          const int32_t mc = (lds_seq[j - 1] == c.letters[node]) ? c.m : c.x;
-          const uint32_t nin = c.in_cnt[node];
-          if (nin == 0) {
-            if (H_ij == c.matrix[j - 1] + mc) {
-              prev_i = 0;
-              prev_j = j - 1;
-              found = true;
-            }
-          } else {
-            for (uint32_t e = 0; e < nin && !found; ++e) {
-              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
-              if (H_ij == c.matrix[static_cast<size_t>(p) * c.MW + j - 1] + mc) {
-                prev_i = p;
-                prev_j = j - 1;
-                found = true;
-              }
-            }
-          }
-        }
-        if (!found && i != 0) {
-          const uint16_t node = c.sorted[i - 1];
-          const uint32_t nin = c.in_cnt[node];
-          if (nin == 0) {
-            if (H_ij == c.matrix[j] + c.g) {
-              prev_i = 0;
-              found = true;
-            }
-          } else {
-            for (uint32_t e = 0; e < nin && !found; ++e) {
-              uint32_t p = c.rank[c.in_edges[node * c.ME + e]] + 1;
-              if (H_ij == c.matrix[static_cast<size_t>(p) * c.MW + j] + c.g) {
-                prev_i = p;
-                found = true;
-              }
-            }
-          }
-        }
-        if (!found && j != 0) {
-          if (H_ij == c.matrix[static_cast<size_t>(i) * c.MW + j - 1] + c.g) {
+        int32_t rec_node = -1, rec_seq = -1;
+        if (i == 0) {
+          // row 0: all-gap prefix — consume remaining columns
+          rec_seq = static_cast<int32_t>(j - 1);
+          prev_j = j - 1;
+        } else {
+          const uint8_t mv = c.moves[static_cast<size_t>(i) * c.MW + j];
+          const uint8_t type = mv & 3;
+          const uint32_t e = mv >> 2;
+          const uint16_t node = s.sorted[i - 1];
+          if (type == kMvLeft) {
+            rec_seq = static_cast<int32_t>(j - 1);
             prev_j = j - 1;
-            found = true;
+          } else {
+            const uint32_t nin = s.in_cnt[node];
+            const uint32_t p = (nin == 0) ? 0u : s.rank[in_edge_of(c, s, node, e)] + 1;
+            if (type == kMvDiag) {
+              rec_node = static_cast<int32_t>(node);
+              rec_seq = static_cast<int32_t>(j - 1);
+              prev_i = p;
+              prev_j = j - 1;
+            } else if (type == kMvUp) {
+              rec_node = static_cast<int32_t>(node);
+              prev_i = p;
+            } else {
+              c.status = kPoaConsensusOverflow;  // invalid move: fail window
+              break;
+            }
           }
         }
-        if (!found) {
-          c.status = kPoaConsensusOverflow;  // inconsistent DP: fail window
-          break;
-        }
-        c.aln_nodes[aln_len] = (i == prev_i) ? -1 : static_cast<int32_t>(c.sorted[i - 1]);
-        c.aln_seq[aln_len] = (j == prev_j) ? -1 : static_cast<int32_t>(j - 1);
+        c.aln_nodes[aln_len] = rec_node;
+        c.aln_seq[aln_len] = rec_seq;
         ++aln_len;
         i = prev_i;
         j = prev_j;
       }
 
       if (c.status == kPoaOk) {
-        add_alignment_d(c, seq, wts, len, aln_len);
+        add_alignment_d(c, s, seq, wts, len, aln_len);
       }
       if (c.status == kPoaOk) {
-        topo_sort_d(c);
+        topo_sort_d(c, s);
       }
     }
 
-    // lane 0's graph updates must be visible to the whole wave
-    __threadfence_block();
+    // lane 0's LDS/graph updates must be visible to the whole wave
+    __syncthreads();
     // broadcast updated scalars from lane 0 to the wave
     c.num_nodes = __shfl(c.num_nodes, 0, kLanes);
     c.seqs_in_graph = __shfl(c.seqs_in_graph, 0, kLanes);
@@ -634,7 +697,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     uint16_t* cov = a.coverage + static_cast<size_t>(win) * L.max_consensus;
     int32_t clen = -1;
     if (c.status == kPoaOk) {
-      clen = consensus_d(c, out, cov, L.max_consensus);
+      clen = consensus_d(c, s, out, cov, L.max_consensus);
     }
     a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
     a.status[win] = c.status;

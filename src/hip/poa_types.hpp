@@ -69,6 +69,9 @@ struct PoaDeviceArena {
   int32_t* aln_nodes;    // [2 * matrix_width + max_nodes] alignment node ids
   int32_t* aln_seq;      // [same] alignment sequence positions
   int16_t* matrix;       // [(max_nodes + 1) * matrix_width] DP scores
+  uint8_t* moves;        // [(max_nodes + 1) * matrix_width] DP move bytes:
+                         // bits 0-1 {0=diag,1=up,2=left,3=invalid}, bits 2-7
+                         // in-edge index of the chosen predecessor
 
   // outputs (D2H once per batch)
   uint8_t* consensus;     // [max_consensus] per window, reversed on host
