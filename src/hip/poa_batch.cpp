@@ -24,6 +24,7 @@ size_t slab_bytes(const PoaLimits& L) {
   b += (2 * L.matrix_width + n) * 8;  // aln_nodes + aln_seq
   b += (n + 1) * L.matrix_width * 2;  // matrix
   b += (n + 1) * L.matrix_width;      // moves
+  b += n * 8;                         // row_desc
   return b;
 }
 
@@ -106,6 +107,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   size_t o_aln_s = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
   size_t o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
   size_t o_moves = carve(num_slabs_ * (n + 1) * L.matrix_width);
+  size_t o_rd = carve(num_slabs_ * n * 8);
   size_t o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
   size_t o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
   size_t o_clen = carve(num_slabs_ * 4);
@@ -136,6 +138,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.aln_seq = reinterpret_cast<int32_t*>(base + o_aln_s);
   arena_.matrix = reinterpret_cast<int16_t*>(base + o_matrix);
   arena_.moves = base + o_moves;
+  arena_.row_desc = reinterpret_cast<uint64_t*>(base + o_rd);
   arena_.consensus = base + o_cons;
   arena_.coverage = reinterpret_cast<uint16_t*>(base + o_cov);
   arena_.consensus_len = reinterpret_cast<uint32_t*>(base + o_clen);
@@ -255,6 +258,29 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   }
 
   pack();
+  const size_t nw0 = windows_.size();
+  // dispatch heaviest windows first: blocks launch roughly in order, so the
+  // long poles start early instead of defining the tail of the launch
+  std::vector<uint32_t> perm(nw0);
+  for (uint32_t i = 0; i < nw0; ++i) perm[i] = i;
+  auto cost = [&](uint32_t w) {
+    const uint32_t first = h_layer_index_[w];
+    return h_layer_ends_[first + h_desc_[w].num_seqs - 1];  // total layer bytes
+  };
+  std::sort(perm.begin(), perm.end(), [&](uint32_t a, uint32_t b) {
+    const uint32_t ca = cost(a), cb = cost(b);
+    if (ca != cb) return ca > cb;
+    return a < b;
+  });
+  std::vector<PoaWindowDesc> desc_sorted(nw0);
+  std::vector<uint32_t> lidx_sorted(nw0);
+  for (uint32_t w = 0; w < nw0; ++w) {
+    desc_sorted[w] = h_desc_[perm[w]];
+    lidx_sorted[w] = h_layer_index_[perm[w]];
+  }
+  std::copy(desc_sorted.begin(), desc_sorted.end(), h_desc_);
+  std::copy(lidx_sorted.begin(), lidx_sorted.end(), h_layer_index_);
+
   RGA_HIP_CHECK(hipSetDevice(device_));
   auto s = static_cast<hipStream_t>(stream_);
   auto d = [&](const void* dst, const void* src, size_t bytes) {
@@ -278,11 +304,13 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, nw * 4, hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipStreamSynchronize(s));
 
-  // CPU-parity post-processing (reference cudabatch.cpp:199-261)
+  // CPU-parity post-processing (reference cudabatch.cpp:199-261).
+  // Kernel slot i handled original window perm[i].
   for (size_t i = 0; i < nw; ++i) {
-    auto& window = windows_[i];
+    const uint32_t orig = perm[i];
+    auto& window = windows_[orig];
     if (h_status_[i] != kPoaOk || h_consensus_len_[i] == 0) {
-      polished[i] = false;  // device-side failure -> CPU fallback
+      polished[orig] = false;  // device-side failure -> CPU fallback
       continue;
     }
     std::string consensus(reinterpret_cast<char*>(h_consensus_ + i * limits_.max_consensus),
@@ -290,7 +318,7 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     bool status = true;
     if (window->type() == WindowType::kTGS && trim) {
       const uint16_t* cov = h_coverage_ + i * limits_.max_consensus;
-      uint32_t average = seqs_added_[i] / 2;
+      uint32_t average = seqs_added_[orig] / 2;
       int32_t begin = 0, end = static_cast<int32_t>(consensus.size()) - 1;
       for (; begin < static_cast<int32_t>(consensus.size()); ++begin) {
         if (cov[begin] >= average) {
@@ -313,7 +341,7 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     if (status) {
       window->set_consensus(std::move(consensus));
     }
-    polished[i] = status;
+    polished[orig] = status;
   }
   return polished;
 }

@@ -1,4 +1,4 @@
-// HIP/CDNA4 POA mega-kernel v2: one 64-lane wavefront per window, LDS-centric.
+// HIP/CDNA4 POA mega-kernel v3: one 64-lane wavefront per window.
 //
 // Per window it loops over the layers: Needleman-Wunsch of the layer against
 // the current POA graph (rows = graph nodes in topological order, columns =
@@ -7,13 +7,19 @@
 // topological re-sort, and finally heaviest-bundle consensus with per-base
 // coverage.
 //
-// v2 vs v1 (v1 streamed every DP row and all graph arrays through HBM and
-// measured 85% of wave-cycles parked on memory waits): the hot graph arrays
-// (letters / rank / sorted / in- & out-counts / first edges) and a 4-row DP
-// ring live in LDS (~31 KB/block, 5 blocks/CU); predecessor rows are LDS
-// hits unless the rank distance exceeds the ring (rare bubbles -> global
-// matrix). The DP records a move byte (type + in-edge index) per cell so the
-// serial traceback is one byte load per step instead of score re-derivation.
+// Perf lineage. v1 streamed every DP value through HBM: 85% of wave-cycles
+// parked on memory (measured, profiles/). v2 mirrored the graph arrays into
+// 31 KB of LDS, which fixed the DP stalls but capped co-residency at 5
+// blocks/CU — the still-global serial phases could no longer be hidden and
+// wall time got worse. v3 keeps only the DP essentials in LDS (4-row ring +
+// layer bases, 9 KB -> ~16 blocks/CU): predecessor rows are LDS hits unless
+// the rank distance exceeds the ring (rare bubbles -> global matrix), the
+// per-row scalar pointer-chase (sorted -> letter/in-count/first-edge ->
+// pred rank) is replaced by one packed u64 `row_desc` per rank built
+// LANE-PARALLEL after each topological sort, and the DP records a move byte
+// per cell so the serial traceback is one byte load per step. The serial
+// graph phases stay in global memory and are hidden by the deep window
+// co-residency, as in v1.
 //
 // Semantics mirror the CPU engine (src/align/poa.cpp) so GPU results are
 // deterministic and window-content-only (reference racon-gpu pins separate
@@ -28,9 +34,9 @@ namespace {
 
 constexpr int kLanes = 64;
 constexpr int32_t kNegInf = -(1 << 28);
-constexpr uint32_t kMaxN = 2048;   // LDS mirrors; PoaLimits.max_nodes must fit
-constexpr uint32_t kMaxW = 1024;   // LDS row width; matrix_width must fit
-constexpr uint32_t kRing = 4;      // DP rows kept in LDS
+constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
+constexpr uint32_t kRing = 4;     // DP rows kept in LDS
+constexpr uint32_t kMaxPre = 8;   // predecessor rows precomputed per row
 
 // move byte encoding
 constexpr uint8_t kMvDiag = 0;
@@ -49,39 +55,34 @@ __device__ inline int32_t wave_scan_max(int32_t v, int lane) {
   return v;
 }
 
-// LDS mirrors of the hot graph arrays + DP row ring. `work` (Kahn scratch)
-// aliases the DP ring: the ring's contents are only read within one layer's
-// DP (every predecessor row is recomputed before use), so the sort may
-// clobber it freely between layers.
-struct Shared {
-  union {
-    int16_t ring[kRing][kMaxW];  // DP rows r-3..r (slot = row % kRing)
-    uint16_t work[kMaxN];        // Kahn in-degree scratch
-  } u;
-  uint16_t rank[kMaxN];
-  uint16_t sorted[kMaxN];
-  uint16_t first_in[kMaxN];
-  uint16_t first_out[kMaxN];
-  uint8_t letters[kMaxN];
-  uint8_t in_cnt[kMaxN];
-  uint8_t out_cnt[kMaxN];
-  uint8_t seq[kMaxW];
-};
+// packed row descriptor (one per topological rank)
+__device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
+                                   uint16_t pred_row, uint8_t is_end) {
+  return static_cast<uint64_t>(letter) | (static_cast<uint64_t>(nin) << 8) |
+         (static_cast<uint64_t>(node) << 16) | (static_cast<uint64_t>(pred_row) << 32) |
+         (static_cast<uint64_t>(is_end) << 48);
+}
 
 struct WindowCtx {
-  // global slab pointers for this window (cold arrays)
+  uint8_t* letters;
+  uint8_t* in_cnt;
+  uint8_t* out_cnt;
+  uint8_t* ring_cnt;
   uint16_t* in_edges;
   int32_t* in_weights;
   uint16_t* out_edges;
-  uint8_t* ring_cnt;
   uint16_t* ring;
   uint16_t* nseq;
+  uint16_t* sorted;
+  uint16_t* rank;
+  uint16_t* work;
   int64_t* hb_score;
   int32_t* hb_pred;
   int32_t* aln_nodes;
   int32_t* aln_seq;
   int16_t* matrix;
   uint8_t* moves;
+  uint64_t* row_desc;
 
   const uint8_t* seq_base;
   const uint8_t* weight_base;
@@ -101,23 +102,13 @@ struct WindowCtx {
 
 // ---------- serial (lane 0) graph helpers ----------
 
-__device__ inline uint16_t in_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
-                                      uint32_t e) {
-  return (e == 0) ? s.first_in[node] : c.in_edges[node * c.ME + e];
-}
-
-__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
-                                       uint32_t e) {
-  return (e == 0) ? s.first_out[node] : c.out_edges[node * c.ME + e];
-}
-
-__device__ bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
-  uint32_t n_out = s.out_cnt[a];
+__device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t w) {
+  uint32_t n_out = c.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
-    if (out_edge_of(c, s, a, e) == b) {
-      uint32_t n_in = s.in_cnt[b];
+    if (c.out_edges[a * c.ME + e] == b) {
+      uint32_t n_in = c.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
-        if (in_edge_of(c, s, b, f) == a) {
+        if (c.in_edges[b * c.ME + f] == a) {
           c.in_weights[b * c.ME + f] += w;
           return true;
         }
@@ -125,34 +116,28 @@ __device__ bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int3
       return true;  // unreachable for a consistent graph
     }
   }
-  if (n_out >= c.ME || s.in_cnt[b] >= c.ME) {
+  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
     c.status = kPoaEdgeOverflow;
     return false;
   }
   c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
-  if (n_out == 0) {
-    s.first_out[a] = static_cast<uint16_t>(b);
-  }
-  s.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
-  uint32_t n_in = s.in_cnt[b];
+  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = c.in_cnt[b];
   c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
-  if (n_in == 0) {
-    s.first_in[b] = static_cast<uint16_t>(a);
-  }
   c.in_weights[b * c.ME + n_in] = w;
-  s.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
-  if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
+__device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
+  if (c.num_nodes >= c.MN) {
     c.status = kPoaNodeOverflow;
     return -1;
   }
   uint32_t id = c.num_nodes++;
-  s.letters[id] = letter;
-  s.in_cnt[id] = 0;
-  s.out_cnt[id] = 0;
+  c.letters[id] = letter;
+  c.in_cnt[id] = 0;
+  c.out_cnt[id] = 0;
   c.ring_cnt[id] = 0;
   c.nseq[id] = 0;
   return static_cast<int32_t>(id);
@@ -160,7 +145,7 @@ __device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
 
 // Threads the traceback path (stored reversed in aln_*) into the graph.
 // Mirrors Graph::add_alignment (src/align/poa.cpp).
-__device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, const uint8_t* wts,
+__device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t* wts,
                                 uint32_t len, int32_t aln_len) {
   // first/last aligned sequence positions
   int32_t first_pos = -1, last_pos = -1;
@@ -178,7 +163,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, con
   int32_t last_counted = -1;
 
   auto link = [&](int32_t a, int32_t b, int32_t w) {
-    if (!add_edge_d(c, s, a, b, w)) {
+    if (!add_edge_d(c, a, b, w)) {
       return;
     }
     if (a != last_counted) {
@@ -195,7 +180,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, con
 
   // head chain: seq[0 .. first_pos)
   for (int32_t p = 0; p < first_pos; ++p) {
-    int32_t id = add_node_d(c, s, seq[p]);
+    int32_t id = add_node_d(c, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -214,22 +199,22 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, con
     int32_t node = c.aln_nodes[k];
     int32_t new_id;
     if (node == -1) {
-      new_id = add_node_d(c, s, letter);
+      new_id = add_node_d(c, letter);
       if (new_id < 0) return;
-    } else if (s.letters[node] == letter) {
+    } else if (c.letters[node] == letter) {
       new_id = node;
     } else {
       new_id = -1;
       uint32_t nr = c.ring_cnt[node];
       for (uint32_t r = 0; r < nr; ++r) {
         uint16_t aid = c.ring[node * c.MR + r];
-        if (s.letters[aid] == letter) {
+        if (c.letters[aid] == letter) {
           new_id = aid;
           break;
         }
       }
       if (new_id == -1) {
-        new_id = add_node_d(c, s, letter);
+        new_id = add_node_d(c, letter);
         if (new_id < 0) return;
         // join the ring: new node linked with node and all its partners
         if (nr >= c.MR) {
@@ -263,7 +248,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, con
 
   // tail chain: seq[last_pos+1 .. len)
   for (int32_t p = (last_pos == -1 ? len : last_pos + 1); p < static_cast<int32_t>(len); ++p) {
-    int32_t id = add_node_d(c, s, seq[p]);
+    int32_t id = add_node_d(c, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -275,49 +260,48 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq, con
   ++c.seqs_in_graph;
 }
 
-// Kahn topological sort (FIFO, deterministic) over the LDS mirrors.
-__device__ void topo_sort_d(WindowCtx& c, Shared& s) {
+// Kahn topological sort (FIFO, deterministic). Rebuilds sorted/rank.
+__device__ void topo_sort_d(WindowCtx& c) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
-    s.u.work[i] = s.in_cnt[i];
+    c.work[i] = c.in_cnt[i];
   }
   uint32_t qhead = 0, qtail = 0;
   for (uint32_t i = 0; i < n; ++i) {
-    if (s.u.work[i] == 0) {
-      s.sorted[qtail++] = static_cast<uint16_t>(i);
+    if (c.work[i] == 0) {
+      c.sorted[qtail++] = static_cast<uint16_t>(i);
     }
   }
   while (qhead < qtail) {
-    uint16_t u = s.sorted[qhead++];
-    uint32_t nout = s.out_cnt[u];
+    uint16_t u = c.sorted[qhead++];
+    uint32_t nout = c.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t v = out_edge_of(c, s, u, e);
-      if (--s.u.work[v] == 0) {
-        s.sorted[qtail++] = v;
+      uint16_t v = c.out_edges[u * c.ME + e];
+      if (--c.work[v] == 0) {
+        c.sorted[qtail++] = v;
       }
     }
   }
   for (uint32_t r = 0; r < qtail; ++r) {
-    s.rank[s.sorted[r]] = static_cast<uint16_t>(r);
+    c.rank[c.sorted[r]] = static_cast<uint16_t>(r);
   }
 }
 
 // Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
 // Returns consensus length written into out/cov (forward order), or -1.
-__device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* cov,
-                               uint32_t max_out) {
+__device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32_t max_out) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
     c.hb_score[i] = -1;
     c.hb_pred[i] = -1;
   }
 
-  uint32_t max_id = s.sorted[0];
+  uint32_t max_id = c.sorted[0];
   for (uint32_t r = 0; r < n; ++r) {
-    uint16_t nid = s.sorted[r];
-    uint32_t nin = s.in_cnt[nid];
+    uint16_t nid = c.sorted[r];
+    uint32_t nin = c.in_cnt[nid];
     for (uint32_t e = 0; e < nin; ++e) {
-      uint16_t p = in_edge_of(c, s, nid, e);
+      uint16_t p = c.in_edges[nid * c.ME + e];
       int64_t w = c.in_weights[nid * c.ME + e];
       if (c.hb_score[nid] < w ||
           (c.hb_score[nid] == w && c.hb_pred[nid] != -1 &&
@@ -339,20 +323,20 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   // otherwise spin forever — those windows fail over to the CPU instead)
   uint32_t guard = 0;
   uint32_t prev_rank = 0;
-  while (s.out_cnt[max_id] != 0) {
-    if (++guard > n || (guard > 1 && s.rank[max_id] <= prev_rank)) {
+  while (c.out_cnt[max_id] != 0) {
+    if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
       c.status = kPoaConsensusOverflow;
       return -1;
     }
-    prev_rank = s.rank[max_id];
-    uint32_t rank0 = s.rank[max_id];
+    prev_rank = c.rank[max_id];
+    uint32_t rank0 = c.rank[max_id];
     // invalidate alternative branches
-    uint32_t nout = s.out_cnt[max_id];
+    uint32_t nout = c.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t endn = out_edge_of(c, s, max_id, e);
-      uint32_t nin = s.in_cnt[endn];
+      uint16_t endn = c.out_edges[max_id * c.ME + e];
+      uint32_t nin = c.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
-        uint16_t o = in_edge_of(c, s, endn, f);
+        uint16_t o = c.in_edges[endn * c.ME + f];
         if (o != max_id) {
           c.hb_score[o] = -1;
         }
@@ -361,12 +345,12 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
     int64_t best = 0;
     uint32_t best_id = 0;
     for (uint32_t r = rank0 + 1; r < n; ++r) {
-      uint16_t nid = s.sorted[r];
+      uint16_t nid = c.sorted[r];
       c.hb_score[nid] = -1;
       c.hb_pred[nid] = -1;
-      uint32_t nin = s.in_cnt[nid];
+      uint32_t nin = c.in_cnt[nid];
       for (uint32_t e = 0; e < nin; ++e) {
-        uint16_t p = in_edge_of(c, s, nid, e);
+        uint16_t p = c.in_edges[nid * c.ME + e];
         if (c.hb_score[p] == -1) {
           continue;
         }
@@ -403,7 +387,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   idw = static_cast<int32_t>(max_id);
   for (int32_t k = path_len - 1; k >= 0; --k) {
     uint32_t node = static_cast<uint32_t>(idw);
-    out[k] = s.letters[node];
+    out[k] = c.letters[node];
     uint32_t covv = c.nseq[node];
     uint32_t nr = c.ring_cnt[node];
     for (uint32_t r = 0; r < nr; ++r) {
@@ -413,6 +397,22 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
     idw = c.hb_pred[idw];
   }
   return path_len;
+}
+
+// Lane-parallel: rebuild the packed per-rank row descriptors from the graph
+// arrays (called after backbone init and after every topo sort).
+__device__ void build_row_desc(WindowCtx& c, int lane) {
+  const uint32_t n = c.num_nodes;
+  for (uint32_t node = lane; node < n; node += kLanes) {
+    const uint32_t r = c.rank[node];
+    const uint8_t nin = c.in_cnt[node];
+    uint16_t pred_row = 0;
+    if (nin > 0) {
+      pred_row = static_cast<uint16_t>(c.rank[c.in_edges[node * c.ME]] + 1);
+    }
+    const uint8_t is_end = (c.out_cnt[node] == 0) ? 1 : 0;
+    c.row_desc[r] = pack_rd(c.letters[node], nin, static_cast<uint16_t>(node), pred_row, is_end);
+  }
 }
 
 // ---------- the mega-kernel ----------
@@ -428,21 +428,29 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const PoaLimits L = a.limits;
   const uint32_t slab = desc.scratch_idx;
 
-  __shared__ Shared s;
+  __shared__ int16_t lds_ring[kRing][kMaxW];
+  __shared__ uint8_t lds_seq[kMaxW];
 
   WindowCtx c;
+  c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
+  c.in_cnt = a.in_cnt + static_cast<size_t>(slab) * L.max_nodes;
+  c.out_cnt = a.out_cnt + static_cast<size_t>(slab) * L.max_nodes;
+  c.ring_cnt = a.ring_cnt + static_cast<size_t>(slab) * L.max_nodes;
   c.in_edges = a.in_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
   c.in_weights = a.in_weights + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
   c.out_edges = a.out_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
-  c.ring_cnt = a.ring_cnt + static_cast<size_t>(slab) * L.max_nodes;
   c.ring = a.ring + static_cast<size_t>(slab) * L.max_nodes * L.max_ring;
   c.nseq = a.nseq + static_cast<size_t>(slab) * L.max_nodes;
+  c.sorted = a.sorted + static_cast<size_t>(slab) * L.max_nodes;
+  c.rank = a.rank + static_cast<size_t>(slab) * L.max_nodes;
+  c.work = a.work + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_score = a.hb_score + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_pred = a.hb_pred + static_cast<size_t>(slab) * L.max_nodes;
   c.aln_nodes = a.aln_nodes + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
   c.aln_seq = a.aln_seq + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);
   c.matrix = a.matrix + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
   c.moves = a.moves + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
+  c.row_desc = a.row_desc + static_cast<size_t>(slab) * L.max_nodes;
 
   c.seq_base = a.seq_data + desc.seq_offset;
   c.weight_base = a.weight_data + desc.seq_offset;
@@ -451,7 +459,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.ME = L.max_edges;
   c.MR = L.max_ring;
   c.MW = L.matrix_width;
-  c.MN = min(L.max_nodes, kMaxN);
+  c.MN = L.max_nodes;
   c.m = a.match;
   c.x = a.mismatch;
   c.g = a.gap;
@@ -462,30 +470,34 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const uint8_t* bb_seq = c.seq_base;
   const uint8_t* bb_wts = c.weight_base;
   for (uint32_t i = lane; i < bb_len; i += kLanes) {
-    s.letters[i] = bb_seq[i];
+    c.letters[i] = bb_seq[i];
     c.ring_cnt[i] = 0;
     c.nseq[i] = bb_len >= 2 ? 1 : 0;
-    s.sorted[i] = static_cast<uint16_t>(i);
-    s.rank[i] = static_cast<uint16_t>(i);
+    c.sorted[i] = static_cast<uint16_t>(i);
+    c.rank[i] = static_cast<uint16_t>(i);
+    uint8_t nin = 0;
     if (i == 0) {
-      s.in_cnt[i] = 0;
+      c.in_cnt[i] = 0;
     } else {
-      s.in_cnt[i] = 1;
-      s.first_in[i] = static_cast<uint16_t>(i - 1);
+      nin = 1;
+      c.in_cnt[i] = 1;
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
+    uint8_t is_end = 0;
     if (i + 1 < bb_len) {
-      s.out_cnt[i] = 1;
-      s.first_out[i] = static_cast<uint16_t>(i + 1);
+      c.out_cnt[i] = 1;
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
-      s.out_cnt[i] = 0;
+      c.out_cnt[i] = 0;
+      is_end = 1;
     }
+    c.row_desc[i] = pack_rd(bb_seq[i], nin, static_cast<uint16_t>(i),
+                            static_cast<uint16_t>(i), is_end);
   }
   c.num_nodes = bb_len;
   c.seqs_in_graph = 1;
-  __syncthreads();  // LDS graph writes -> visible to all lanes
+  __syncthreads();  // graph writes -> visible to all lanes
 
   // ---- per-layer loop ----
   for (uint32_t layer = 1; layer < c.num_seqs && c.status == kPoaOk; ++layer) {
@@ -498,7 +510,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     }
 
     for (uint32_t j = lane; j < len; j += kLanes) {
-      s.seq[j] = seq[j];
+      lds_seq[j] = seq[j];
     }
     __syncthreads();
 
@@ -511,26 +523,48 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
 
     // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
     for (uint32_t r = 0; r < n; ++r) {
-      const uint16_t node = s.sorted[r];
-      const uint8_t letter = s.letters[node];
-      const uint32_t nin = s.in_cnt[node];
+      const uint64_t rd = c.row_desc[r];
+      const uint8_t letter = static_cast<uint8_t>(rd);
+      const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
+      const uint32_t node = static_cast<uint32_t>((rd >> 16) & 0xffff);
+      const uint32_t pred0 = static_cast<uint32_t>((rd >> 32) & 0xffff);
+      const bool is_end = ((rd >> 48) & 0xff) != 0;
       int16_t* Hrow = c.matrix + static_cast<size_t>(r + 1) * c.MW;
       uint8_t* Mrow = c.moves + static_cast<size_t>(r + 1) * c.MW;
-      int16_t* ring_row = s.u.ring[(r + 1) % kRing];
+      int16_t* ring_row = lds_ring[(r + 1) % kRing];
+
+      // predecessor rows (e < kMaxPre precomputed; beyond that re-derived
+      // in the chunk loop — nodes with >8 in-edges are rare)
+      uint32_t pred_rows[kMaxPre];
+      pred_rows[0] = pred0;
+      const uint32_t npre = min(nin, kMaxPre);
+      for (uint32_t e = 1; e < npre; ++e) {
+        pred_rows[e] = c.rank[c.in_edges[node * c.ME + e]] + 1;
+      }
+
+      // fetch a predecessor row value: LDS ring if close, global otherwise
+      auto pred_val = [&](uint32_t p, uint32_t col) -> int32_t {
+        if (p == 0) {
+          return static_cast<int32_t>(col) * c.g;  // arithmetic row 0
+        }
+        if (r + 1 - p < kRing) {
+          return lds_ring[p % kRing][col];
+        }
+        return c.matrix[static_cast<size_t>(p) * c.MW + col];
+      };
 
       // first column (j = 0): max over preds of Hp[0] + gap
       int32_t h0;
+      uint32_t e0 = 0;
       {
         int32_t best0 = kNegInf;
-        uint32_t e0 = 0;
         if (nin == 0) {
           best0 = 0;
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
-            const uint32_t p = s.rank[in_edge_of(c, s, node, e)] + 1;
-            const int32_t hp0 = (r + 1 - p < kRing)
-                                    ? s.u.ring[p % kRing][0]
-                                    : c.matrix[static_cast<size_t>(p) * c.MW];
+            const uint32_t p = (e < kMaxPre) ? pred_rows[e]
+                                             : c.rank[c.in_edges[node * c.ME + e]] + 1;
+            const int32_t hp0 = pred_val(p, 0);
             if (hp0 > best0) {
               best0 = hp0;
               e0 = e;
@@ -556,7 +590,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         uint32_t ve = 0;
         uint8_t vtype = kMvInvalid;
         if (j < width) {
-          const int32_t sub = (s.seq[j - 1] == letter) ? c.m : c.x;
+          const int32_t sub = (lds_seq[j - 1] == letter) ? c.m : c.x;
           if (nin == 0) {
             // pred is the arithmetic row 0
             const int32_t diag = static_cast<int32_t>(j - 1) * c.g + sub;
@@ -567,17 +601,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
             int32_t best_diag = kNegInf, best_up = kNegInf;
             uint32_t e_diag = 0, e_up = 0;
             for (uint32_t e = 0; e < nin; ++e) {
-              const uint32_t p = s.rank[in_edge_of(c, s, node, e)] + 1;
-              int32_t hpjm1, hpj;
-              if (r + 1 - p < kRing) {
-                const int16_t* Rp = s.u.ring[p % kRing];
-                hpjm1 = Rp[j - 1];
-                hpj = Rp[j];
-              } else {
-                const int16_t* Hp = c.matrix + static_cast<size_t>(p) * c.MW;
-                hpjm1 = Hp[j - 1];
-                hpj = Hp[j];
-              }
+              const uint32_t p = (e < kMaxPre) ? pred_rows[e]
+                                               : c.rank[c.in_edges[node * c.ME + e]] + 1;
+              const int32_t hpjm1 = pred_val(p, j - 1);
+              const int32_t hpj = pred_val(p, j);
               if (hpjm1 + sub > best_diag) {
                 best_diag = hpjm1 + sub;
                 e_diag = e;
@@ -616,12 +643,11 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         carry_u = __shfl(u, kLanes - 1, kLanes);
       }
 
-      // ring/global row writes must be visible to every lane before the next
-      // row (cross-lane read-after-write within one wave)
+      // ring writes must be visible to every lane before the next row
       __syncthreads();
 
       // end-node max (strict >, first in topological order wins)
-      if (s.out_cnt[node] == 0) {
+      if (is_end) {
         const int src_lane = static_cast<int>((len - 1) % kLanes);
         const int32_t lc = __shfl(last_col_val, src_lane, kLanes);
         if (lc > best_score) {
@@ -647,13 +673,18 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           const uint8_t mv = c.moves[static_cast<size_t>(i) * c.MW + j];
           const uint8_t type = mv & 3;
           const uint32_t e = mv >> 2;
-          const uint16_t node = s.sorted[i - 1];
+          const uint64_t rd = c.row_desc[i - 1];
+          const uint32_t node = static_cast<uint32_t>((rd >> 16) & 0xffff);
+          const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
           if (type == kMvLeft) {
             rec_seq = static_cast<int32_t>(j - 1);
             prev_j = j - 1;
           } else {
-            const uint32_t nin = s.in_cnt[node];
-            const uint32_t p = (nin == 0) ? 0u : s.rank[in_edge_of(c, s, node, e)] + 1;
+            uint32_t p = 0;
+            if (nin != 0) {
+              p = (e == 0) ? static_cast<uint32_t>((rd >> 32) & 0xffff)
+                           : c.rank[c.in_edges[node * c.ME + e]] + 1;
+            }
             if (type == kMvDiag) {
               rec_node = static_cast<int32_t>(node);
               rec_seq = static_cast<int32_t>(j - 1);
@@ -676,19 +707,25 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       }
 
       if (c.status == kPoaOk) {
-        add_alignment_d(c, s, seq, wts, len, aln_len);
+        add_alignment_d(c, seq, wts, len, aln_len);
       }
       if (c.status == kPoaOk) {
-        topo_sort_d(c, s);
+        topo_sort_d(c);
       }
     }
 
-    // lane 0's LDS/graph updates must be visible to the whole wave
+    // lane 0's graph updates must be visible to the whole wave
     __syncthreads();
     // broadcast updated scalars from lane 0 to the wave
     c.num_nodes = __shfl(c.num_nodes, 0, kLanes);
     c.seqs_in_graph = __shfl(c.seqs_in_graph, 0, kLanes);
     c.status = __shfl(c.status, 0, kLanes);
+
+    // rebuild the packed row descriptors for the grown graph, lane-parallel
+    if (c.status == kPoaOk) {
+      build_row_desc(c, lane);
+    }
+    __syncthreads();
   }
 
   // ---- consensus ----
@@ -697,7 +734,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     uint16_t* cov = a.coverage + static_cast<size_t>(win) * L.max_consensus;
     int32_t clen = -1;
     if (c.status == kPoaOk) {
-      clen = consensus_d(c, s, out, cov, L.max_consensus);
+      clen = consensus_d(c, out, cov, L.max_consensus);
     }
     a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
     a.status[win] = c.status;

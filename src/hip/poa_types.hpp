@@ -72,6 +72,9 @@ struct PoaDeviceArena {
   uint8_t* moves;        // [(max_nodes + 1) * matrix_width] DP move bytes:
                          // bits 0-1 {0=diag,1=up,2=left,3=invalid}, bits 2-7
                          // in-edge index of the chosen predecessor
+  uint64_t* row_desc;    // [max_nodes] per-rank packed row descriptor, built
+                         // lane-parallel after each topo sort: letter(8) |
+                         // nin(8) | node(16) | pred_row(16) | is_end(8)
 
   // outputs (D2H once per batch)
   uint8_t* consensus;     // [max_consensus] per window, reversed on host

@@ -1,0 +1,76 @@
+"""CLI negative / argument-validation tests (parity with the reference's
+EXPECT_DEATH set, test/racon_test.cpp:55-86) plus basic CLI behavior."""
+
+import subprocess
+
+import pytest
+
+
+def run_racon(racon_cli, args, **kw):
+    return subprocess.run([racon_cli] + args, capture_output=True, text=True, **kw)
+
+
+def test_window_length_error(racon_cli, tmp_path):
+    reads = tmp_path / "r.fasta"
+    reads.write_text(">a\nACGT\n")
+    out = run_racon(racon_cli, ["-w", "0", str(reads), str(reads), str(reads)])
+    assert out.returncode != 0
+    assert "invalid window length" in out.stderr
+
+
+def test_sequences_extension_error(racon_cli, tmp_path):
+    bad = tmp_path / "reads.txt"
+    bad.write_text("x")
+    out = run_racon(racon_cli, [str(bad), str(bad), str(bad)])
+    assert out.returncode != 0
+    assert "unsupported format extension" in out.stderr
+    assert ".fasta" in out.stderr
+
+
+def test_overlaps_extension_error(racon_cli, tmp_path):
+    reads = tmp_path / "r.fasta"
+    reads.write_text(">a\nACGT\n")
+    bad = tmp_path / "ovl.txt"
+    bad.write_text("x")
+    out = run_racon(racon_cli, [str(reads), str(bad), str(reads)])
+    assert out.returncode != 0
+    assert "unsupported format extension" in out.stderr
+    assert ".paf" in out.stderr
+
+
+def test_target_extension_error(racon_cli, tmp_path):
+    reads = tmp_path / "r.fasta"
+    reads.write_text(">a\nACGT\n")
+    ovl = tmp_path / "o.paf"
+    ovl.write_text("")
+    bad = tmp_path / "t.txt"
+    bad.write_text("x")
+    out = run_racon(racon_cli, [str(reads), str(ovl), str(bad)])
+    assert out.returncode != 0
+    assert "unsupported format extension" in out.stderr
+
+
+def test_missing_positional_arguments(racon_cli):
+    out = run_racon(racon_cli, [])
+    assert out.returncode != 0
+
+
+def test_help_and_version(racon_cli):
+    out = run_racon(racon_cli, ["-h"])
+    assert out.returncode == 0
+    assert "usage: racon" in out.stdout
+    out = run_racon(racon_cli, ["--version"])
+    assert out.returncode == 0
+    assert out.stdout.startswith("v")
+
+
+def test_cli_polishes_sample(racon_cli, sample):
+    out = run_racon(racon_cli, ["-t", "2", sample["reads"], sample["overlaps"],
+                                sample["layout"]])
+    assert out.returncode == 0
+    lines = out.stdout.splitlines()
+    assert lines[0].startswith(">")
+    # output tags are part of the format contract (LN/RC/XC,
+    # reference src/polisher.cpp:522-525)
+    assert "LN:i:" in lines[0] and "RC:i:" in lines[0] and "XC:f:" in lines[0]
+    assert len("".join(lines[1:])) > 10000
