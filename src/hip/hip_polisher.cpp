@@ -23,6 +23,81 @@
 
 namespace rga {
 
+namespace {
+
+// Process-wide batch pools: arena construction (tens of GB of device memory
+// + pinned staging) is expensive, and repeated polish() calls in one process
+// (bench steps, services) would otherwise pay it every time. Batches are
+// keyed by their construction parameters; budget differences are ignored
+// (arenas are internally capped).
+struct BatchPools {
+  std::mutex mutex;
+  std::vector<std::pair<uint64_t, std::unique_ptr<hip::AlignerBatch>>> aligner;
+  std::vector<std::pair<uint64_t, std::unique_ptr<hip::PoaBatch>>> poa;
+};
+BatchPools& pools() {
+  static BatchPools p;
+  return p;
+}
+
+std::unique_ptr<hip::AlignerBatch> acquire_aligner(int device, size_t budget, uint32_t band) {
+  const uint64_t key = (static_cast<uint64_t>(device) << 32) | band;
+  auto& p = pools();
+  {
+    std::lock_guard<std::mutex> lock(p.mutex);
+    for (auto it = p.aligner.begin(); it != p.aligner.end(); ++it) {
+      if (it->first == key) {
+        auto b = std::move(it->second);
+        p.aligner.erase(it);
+        return b;
+      }
+    }
+  }
+  return std::make_unique<hip::AlignerBatch>(device, budget, band);
+}
+
+void release_aligner(int device, uint32_t band, std::unique_ptr<hip::AlignerBatch> b) {
+  const uint64_t key = (static_cast<uint64_t>(device) << 32) | band;
+  b->reset();
+  auto& p = pools();
+  std::lock_guard<std::mutex> lock(p.mutex);
+  p.aligner.emplace_back(key, std::move(b));
+}
+
+uint64_t poa_key(int device, int8_t m, int8_t x, int8_t g, bool banded, uint32_t depth) {
+  return (static_cast<uint64_t>(device) << 40) |
+         (static_cast<uint64_t>(static_cast<uint8_t>(m)) << 32) |
+         (static_cast<uint64_t>(static_cast<uint8_t>(x)) << 24) |
+         (static_cast<uint64_t>(static_cast<uint8_t>(g)) << 16) |
+         (static_cast<uint64_t>(banded) << 15) | depth;
+}
+
+std::unique_ptr<hip::PoaBatch> acquire_poa(int device, size_t budget, int8_t m, int8_t x,
+                                           int8_t g, bool banded, uint32_t depth) {
+  const uint64_t key = poa_key(device, m, x, g, banded, depth);
+  auto& p = pools();
+  {
+    std::lock_guard<std::mutex> lock(p.mutex);
+    for (auto it = p.poa.begin(); it != p.poa.end(); ++it) {
+      if (it->first == key) {
+        auto b = std::move(it->second);
+        p.poa.erase(it);
+        return b;
+      }
+    }
+  }
+  return std::make_unique<hip::PoaBatch>(device, budget, m, x, g, banded, depth);
+}
+
+void release_poa(uint64_t key, std::unique_ptr<hip::PoaBatch> b) {
+  b->reset();
+  auto& p = pools();
+  std::lock_guard<std::mutex> lock(p.mutex);
+  p.poa.emplace_back(key, std::move(b));
+}
+
+}  // namespace
+
 class HipPolisher : public Polisher {
  public:
   HipPolisher(std::unique_ptr<SequenceParser> sparser, std::unique_ptr<OverlapParser> oparser,
@@ -71,13 +146,15 @@ class HipPolisher : public Polisher {
     }
 
     std::vector<std::unique_ptr<hip::AlignerBatch>> batches;
+    std::vector<int> batch_devices;
     for (int d : devices_) {
       RGA_HIP_CHECK(hipSetDevice(d));
       size_t free_mem = 0, total_mem = 0;
       RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
       size_t budget = free_mem * 9 / 10 / config_.aligner_batches;
       for (uint32_t b = 0; b < config_.aligner_batches; ++b) {
-        batches.emplace_back(std::make_unique<hip::AlignerBatch>(d, budget, band));
+        batches.emplace_back(acquire_aligner(d, budget, band));
+        batch_devices.emplace_back(d);
       }
     }
 
@@ -130,6 +207,9 @@ class HipPolisher : public Polisher {
     for (auto& t : threads) {
       t.join();
     }
+    for (size_t b = 0; b < batches.size(); ++b) {
+      release_aligner(batch_devices[b], band, std::move(batches[b]));
+    }
     batches.clear();
 
     fprintf(stderr,
@@ -154,16 +234,19 @@ class HipPolisher : public Polisher {
     logger_->log();
 
     // one batch object per (device, batch) pair, reference cudapolisher.cpp:228-240
+    constexpr uint32_t kMaxDepth = 200;  // MAX_DEPTH_PER_WINDOW, cudapolisher.cpp:226
     std::vector<std::unique_ptr<hip::PoaBatch>> batches;
+    std::vector<uint64_t> batch_keys;
     for (int d : devices_) {
       RGA_HIP_CHECK(hipSetDevice(d));
       size_t free_mem = 0, total_mem = 0;
       RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
       size_t budget = free_mem * 9 / 10 / config_.poa_batches;
       for (uint32_t b = 0; b < config_.poa_batches; ++b) {
-        batches.emplace_back(std::make_unique<hip::PoaBatch>(
-            d, budget, config_.match, config_.mismatch, config_.gap, config_.banded_poa,
-            200 /* MAX_DEPTH_PER_WINDOW, reference cudapolisher.cpp:226 */));
+        batches.emplace_back(acquire_poa(d, budget, config_.match, config_.mismatch,
+                                         config_.gap, config_.banded_poa, kMaxDepth));
+        batch_keys.emplace_back(poa_key(d, config_.match, config_.mismatch, config_.gap,
+                                        config_.banded_poa, kMaxDepth));
       }
     }
 
@@ -228,6 +311,10 @@ class HipPolisher : public Polisher {
     for (auto& t : threads) {
       t.join();
     }
+    for (size_t b = 0; b < batches.size(); ++b) {
+      release_poa(batch_keys[b], std::move(batches[b]));
+    }
+    batches.clear();
     fprintf(stderr,
             "[racon::HipPolisher] poa timings: queue %.3f s, pack+gpu+post %.3f s "
             "(sum over %zu batch threads)\n",
