@@ -44,6 +44,14 @@ constexpr uint8_t kMvUp = 1;
 constexpr uint8_t kMvLeft = 2;
 constexpr uint8_t kMvInvalid = 3;
 
+// Single-wavefront LDS visibility: waits only the LDS (lgkm) counter and
+// stops compiler reordering. Unlike __syncthreads(), it does NOT drain the
+// outstanding global (vm) stores of the row just written — those are
+// fire-and-forget into HBM and must not sit on the per-row critical path.
+__device__ inline void wave_lds_sync() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 __device__ inline int32_t wave_scan_max(int32_t v, int lane) {
   // inclusive max-scan over the 64-lane wavefront
   for (int d = 1; d < kLanes; d <<= 1) {
@@ -522,8 +530,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     uint32_t best_row = 0;
 
     // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
+    uint64_t rd_next = c.row_desc[0];
     for (uint32_t r = 0; r < n; ++r) {
-      const uint64_t rd = c.row_desc[r];
+      const uint64_t rd = rd_next;
+      if (r + 1 < n) {
+        rd_next = c.row_desc[r + 1];  // prefetch: off the critical path
+      }
       const uint8_t letter = static_cast<uint8_t>(rd);
       const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
       const uint32_t node = static_cast<uint32_t>((rd >> 16) & 0xffff);
@@ -643,8 +655,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         carry_u = __shfl(u, kLanes - 1, kLanes);
       }
 
-      // ring writes must be visible to every lane before the next row
-      __syncthreads();
+      // ring writes must be visible to every lane before the next row;
+      // deliberately NOT __syncthreads (would drain the global row stores)
+      wave_lds_sync();
 
       // end-node max (strict >, first in topological order wins)
       if (is_end) {
