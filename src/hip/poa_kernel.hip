@@ -63,12 +63,18 @@ __device__ inline int32_t wave_scan_max(int32_t v, int lane) {
   return v;
 }
 
-// packed row descriptor (one per topological rank)
+// packed row descriptor (one per topological rank); flags: bit0 = end node
+// (no out-edges), bit1 = some successor is beyond the LDS ring, so this
+// row's scores must go to global memory (matrix rows are otherwise
+// write-only traffic nobody reads — the dominant HBM cost of the DP).
+constexpr uint64_t kRdEnd = 1;
+constexpr uint64_t kRdStore = 2;
+
 __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
-                                   uint16_t pred_row, uint8_t is_end) {
+                                   uint16_t pred_row, uint8_t flags) {
   return static_cast<uint64_t>(letter) | (static_cast<uint64_t>(nin) << 8) |
          (static_cast<uint64_t>(node) << 16) | (static_cast<uint64_t>(pred_row) << 32) |
-         (static_cast<uint64_t>(is_end) << 48);
+         (static_cast<uint64_t>(flags) << 48);
 }
 
 struct WindowCtx {
@@ -418,8 +424,16 @@ __device__ void build_row_desc(WindowCtx& c, int lane) {
     if (nin > 0) {
       pred_row = static_cast<uint16_t>(c.rank[c.in_edges[node * c.ME]] + 1);
     }
-    const uint8_t is_end = (c.out_cnt[node] == 0) ? 1 : 0;
-    c.row_desc[r] = pack_rd(c.letters[node], nin, static_cast<uint16_t>(node), pred_row, is_end);
+    const uint32_t nout = c.out_cnt[node];
+    uint8_t flags = (nout == 0) ? kRdEnd : 0;
+    for (uint32_t e = 0; e < nout; ++e) {
+      const uint32_t sr = c.rank[c.out_edges[node * c.ME + e]];
+      if (sr - r >= kRing) {
+        flags |= kRdStore;
+        break;
+      }
+    }
+    c.row_desc[r] = pack_rd(c.letters[node], nin, static_cast<uint16_t>(node), pred_row, flags);
   }
 }
 
@@ -492,16 +506,16 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
-    uint8_t is_end = 0;
+    uint8_t flags = 0;
     if (i + 1 < bb_len) {
       c.out_cnt[i] = 1;
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
       c.out_cnt[i] = 0;
-      is_end = 1;
+      flags = kRdEnd;
     }
     c.row_desc[i] = pack_rd(bb_seq[i], nin, static_cast<uint16_t>(i),
-                            static_cast<uint16_t>(i), is_end);
+                            static_cast<uint16_t>(i), flags);
   }
   c.num_nodes = bb_len;
   c.seqs_in_graph = 1;
@@ -540,7 +554,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
       const uint32_t node = static_cast<uint32_t>((rd >> 16) & 0xffff);
       const uint32_t pred0 = static_cast<uint32_t>((rd >> 32) & 0xffff);
-      const bool is_end = ((rd >> 48) & 0xff) != 0;
+      const uint64_t flags = (rd >> 48) & 0xff;
+      const bool is_end = (flags & kRdEnd) != 0;
+      const bool store_row = (flags & kRdStore) != 0;
       int16_t* Hrow = c.matrix + static_cast<size_t>(r + 1) * c.MW;
       uint8_t* Mrow = c.moves + static_cast<size_t>(r + 1) * c.MW;
       int16_t* ring_row = lds_ring[(r + 1) % kRing];
@@ -585,7 +601,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         }
         h0 = best0 + c.g;
         if (lane == 0) {
-          Hrow[0] = static_cast<int16_t>(h0);
+          if (store_row) {
+            Hrow[0] = static_cast<int16_t>(h0);
+          }
           ring_row[0] = static_cast<int16_t>(h0);
           // column 0 is always a vertical chain through the argmax edge
           Mrow[0] = static_cast<uint8_t>(kMvUp | (e0 << 2));
@@ -644,7 +662,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         u = max(u, carry_u);
         const int32_t h = u + static_cast<int32_t>(j) * c.g;
         if (j < width) {
-          Hrow[j] = static_cast<int16_t>(h);
+          if (store_row) {
+            Hrow[j] = static_cast<int16_t>(h);
+          }
           ring_row[j] = static_cast<int16_t>(h);
           // left move only when the scan strictly beat this cell's v
           Mrow[j] = (h == v) ? static_cast<uint8_t>(vtype | (ve << 2)) : kMvLeft;
