@@ -25,6 +25,7 @@ size_t slab_bytes(const PoaLimits& L) {
   b += (n + 1) * L.matrix_width * 2;  // matrix
   b += (n + 1) * L.matrix_width;      // moves
   b += n * 8;                         // row_desc
+  b += 64;                            // timing
   return b;
 }
 
@@ -108,6 +109,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   size_t o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
   size_t o_moves = carve(num_slabs_ * (n + 1) * L.matrix_width);
   size_t o_rd = carve(num_slabs_ * n * 8);
+  size_t o_timing = carve(static_cast<size_t>(num_slabs_) * 8 * 8);
   size_t o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
   size_t o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
   size_t o_clen = carve(num_slabs_ * 4);
@@ -139,6 +141,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.matrix = reinterpret_cast<int16_t*>(base + o_matrix);
   arena_.moves = base + o_moves;
   arena_.row_desc = reinterpret_cast<uint64_t*>(base + o_rd);
+  arena_.timing = reinterpret_cast<unsigned long long*>(base + o_timing);
   arena_.consensus = base + o_cons;
   arena_.coverage = reinterpret_cast<uint16_t*>(base + o_cov);
   arena_.consensus_len = reinterpret_cast<uint32_t*>(base + o_clen);
@@ -303,6 +306,19 @@ std::vector<bool> PoaBatch::generate(bool trim) {
                                hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, nw * 4, hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipStreamSynchronize(s));
+
+  if (getenv("RGA_POA_TIMING") != nullptr) {
+    std::vector<unsigned long long> t(nw * 8);
+    RGA_HIP_CHECK(hipMemcpy(t.data(), arena_.timing, nw * 8 * 8, hipMemcpyDeviceToHost));
+    unsigned long long sum[8] = {0};
+    for (size_t i = 0; i < nw; ++i) {
+      for (int k = 0; k < 8; ++k) sum[k] += t[i * 8 + k];
+    }
+    fprintf(stderr,
+            "[rga::hip::PoaBatch] timing (wall ticks, %zu windows): dp=%llu tb=%llu "
+            "add=%llu topo=%llu rdesc=%llu cons=%llu total=%llu layers=%llu\n",
+            nw, sum[0], sum[1], sum[2], sum[3], sum[4], sum[5], sum[6], sum[7]);
+  }
 
   // CPU-parity post-processing (reference cudabatch.cpp:199-261).
   // Kernel slot i handled original window perm[i].

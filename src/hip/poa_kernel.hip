@@ -474,6 +474,18 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.moves = a.moves + static_cast<size_t>(slab) * (L.max_nodes + 1) * L.matrix_width;
   c.row_desc = a.row_desc + static_cast<size_t>(slab) * L.max_nodes;
 
+  unsigned long long* timing = a.timing + static_cast<size_t>(win) * 8;
+  unsigned long long tick = wall_clock64();
+  const unsigned long long t_start = tick;
+  unsigned long long t_dp = 0, t_tb = 0, t_add = 0, t_topo = 0, t_rd = 0, t_cons = 0;
+  unsigned long long layers_done = 0;
+  auto lap = [&]() {
+    unsigned long long now = wall_clock64();
+    unsigned long long d = now - tick;
+    tick = now;
+    return d;
+  };
+
   c.seq_base = a.seq_data + desc.seq_offset;
   c.weight_base = a.weight_data + desc.seq_offset;
   c.ends = a.layer_ends + a.layer_ends_index[win];
@@ -542,6 +554,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
 
     int32_t best_score = kNegInf;
     uint32_t best_row = 0;
+    (void)lap();
 
     // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
     uint64_t rd_next = c.row_desc[0];
@@ -690,6 +703,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       }
     }
 
+    t_dp += lap();
+    ++layers_done;
+
     // ---- serial phases on lane 0 ----
     if (lane == 0) {
       // move-byte traceback
@@ -739,12 +755,15 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         j = prev_j;
       }
 
+      t_tb += lap();
       if (c.status == kPoaOk) {
         add_alignment_d(c, seq, wts, len, aln_len);
       }
+      t_add += lap();
       if (c.status == kPoaOk) {
         topo_sort_d(c);
       }
+      t_topo += lap();
     }
 
     // lane 0's graph updates must be visible to the whole wave
@@ -755,10 +774,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     c.status = __shfl(c.status, 0, kLanes);
 
     // rebuild the packed row descriptors for the grown graph, lane-parallel
+    (void)lap();
     if (c.status == kPoaOk) {
       build_row_desc(c, lane);
     }
     __syncthreads();
+    t_rd += lap();
   }
 
   // ---- consensus ----
@@ -766,11 +787,21 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     uint8_t* out = a.consensus + static_cast<size_t>(win) * L.max_consensus;
     uint16_t* cov = a.coverage + static_cast<size_t>(win) * L.max_consensus;
     int32_t clen = -1;
+    (void)lap();
     if (c.status == kPoaOk) {
       clen = consensus_d(c, out, cov, L.max_consensus);
     }
+    t_cons = lap();
     a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
     a.status[win] = c.status;
+    timing[0] = t_dp;
+    timing[1] = t_tb;
+    timing[2] = t_add;
+    timing[3] = t_topo;
+    timing[4] = t_rd;
+    timing[5] = t_cons;
+    timing[6] = wall_clock64() - t_start;
+    timing[7] = layers_done;
   }
 }
 

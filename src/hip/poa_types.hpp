@@ -76,6 +76,12 @@ struct PoaDeviceArena {
                          // lane-parallel after each topo sort: letter(8) |
                          // nin(8) | node(16) | pred_row(16) | is_end(8)
 
+  // per-window phase timing (wall_clock64 deltas; slots: 0 DP rows,
+  // 1 traceback, 2 add_alignment, 3 topo sort, 4 row_desc, 5 consensus,
+  // 6 total, 7 layers processed) — aggregated by the host under
+  // RGA_POA_TIMING for kernel-phase attribution
+  unsigned long long* timing;  // [8] per window
+
   // outputs (D2H once per batch)
   uint8_t* consensus;     // [max_consensus] per window, reversed on host
   uint16_t* coverage;     // [max_consensus] per window
