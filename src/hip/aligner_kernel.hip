@@ -32,10 +32,13 @@ __device__ inline uint32_t base_code(uint8_t b) {
   }
 }
 
-// band top block for column j: clamp(center/64 - K/2, [0, nbt-K])
+// band top block for column j: clamp(center/64 - K/2, [0, nbt-K]).
+// center = (j * n) / m computed as a fixed-point multiply with a
+// precomputed 32.32 step — the integer division was microcoded and ran
+// once per column in the forward pass and twice per traceback step.
 template <int K>
-__device__ inline int32_t btop_of(int64_t j, int64_t n, int64_t m, int32_t nbt) {
-  int32_t center_blk = static_cast<int32_t>((j * n) / m) >> 6;
+__device__ inline int32_t btop_of(uint64_t j, uint64_t step, int32_t nbt) {
+  int32_t center_blk = static_cast<int32_t>((j * step) >> 32) >> 6;
   int32_t top = center_blk - K / 2;
   int32_t hi = nbt > K ? nbt - K : 0;
   return top < 0 ? 0 : (top > hi ? hi : top);
@@ -68,6 +71,9 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
   uint64_t* peq = a.peq + wd.peq_off;
   uint64_t* tb = a.tb + wd.tb_off;
   int32_t* sb = a.sbuf + wd.s_off;
+  // 32.32 fixed-point slope for the band center (one division per lane)
+  const uint64_t step = (m > 0) ? ((static_cast<uint64_t>(n) << 32) / static_cast<uint32_t>(m))
+                                : 0;
 
   // ---- fill Peq (wave-coalesced layout [(b*4+c)*64+lane]) ----
   for (uint32_t b = 0; b < wd.nb; ++b) {
@@ -112,7 +118,7 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
   for (int32_t j = 1; j <= static_cast<int32_t>(wd.mmax); ++j) {
     if (j <= m) {
       const uint32_t c = base_code(t[j - 1]);
-      const int32_t btop_new = btop_of<K>(j, n, m, nbt);
+      const int32_t btop_new = btop_of<K>(j, step, nbt);
       while (btop < btop_new) {
         // band slides down one block: drop top, append pessimistic bottom
 #pragma unroll
@@ -183,8 +189,8 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
   while (st == kAlnOk && i > 0 && j > 0) {
     const int32_t babs = (i - 1) >> 6;
     const uint32_t k = (i - 1) & 63;
-    const int32_t btj = btop_of<K>(j, n, m, nbt);
-    const int32_t btj1 = btop_of<K>(j - 1, n, m, nbt);
+    const int32_t btj = btop_of<K>(j, step, nbt);
+    const int32_t btj1 = btop_of<K>(j - 1, step, nbt);
     const int32_t rbj = babs - btj;
     const int32_t rbj1 = babs - btj1;
     if (rbj < 0 || rbj >= K || rbj1 < 0 || rbj1 >= K) {

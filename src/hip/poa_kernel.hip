@@ -52,14 +52,25 @@ __device__ inline void wave_lds_sync() {
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 }
 
-__device__ inline int32_t wave_scan_max(int32_t v, int lane) {
-  // inclusive max-scan over the 64-lane wavefront
-  for (int d = 1; d < kLanes; d <<= 1) {
-    int32_t o = __shfl_up(v, d, kLanes);
-    if (lane >= d) {
-      v = max(v, o);
-    }
-  }
+// Inclusive max-scan over the 64-lane wavefront via DPP row ops (VALU-speed;
+// the naive __shfl_up chain is 6 dependent ds_bpermute round-trips through
+// the LDS unit and dominated the DP phase — measured with RGA_POA_TIMING).
+// Sequence: shr1/2/4/8 within 16-lane rows, then bcast15 into rows 1&3 and
+// bcast31 into rows 2&3 (the classic GCN prefix idiom with max).
+__device__ inline int32_t wave_scan_max(int32_t v, int /*lane*/) {
+  int32_t t;
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x111, 0xf, 0xf, false);  // row_shr:1
+  v = max(v, t);
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x112, 0xf, 0xf, false);  // row_shr:2
+  v = max(v, t);
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x114, 0xf, 0xf, false);  // row_shr:4
+  v = max(v, t);
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x118, 0xf, 0xf, false);  // row_shr:8
+  v = max(v, t);
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x142, 0xa, 0xf, false);  // row_bcast:15
+  v = max(v, t);
+  t = __builtin_amdgcn_update_dpp(kNegInf, v, 0x143, 0xc, 0xf, false);  // row_bcast:31
+  v = max(v, t);
   return v;
 }
 
@@ -685,7 +696,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
             last_col_val = h;
           }
         }
-        carry_u = __shfl(u, kLanes - 1, kLanes);
+        carry_u = __builtin_amdgcn_readlane(u, kLanes - 1);
       }
 
       // ring writes must be visible to every lane before the next row;

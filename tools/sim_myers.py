@@ -27,7 +27,8 @@ def brute_edit(q, t):
 
 
 def btop_of(j, n, m, nbt, K):
-    center_blk = ((j * n) // m) >> 6
+    step = (n << 32) // m  # 32.32 fixed point, as in the kernel
+    center_blk = (((j * step) >> 32) & 0xFFFFFFFF) >> 6
     top = center_blk - K // 2
     hi = nbt - K if nbt > K else 0
     return max(0, min(top, hi))
