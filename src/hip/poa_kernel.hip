@@ -35,6 +35,7 @@ namespace {
 constexpr int kLanes = 64;
 constexpr int32_t kNegInf = -(1 << 28);
 constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
+constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
 constexpr uint32_t kMaxPre = 8;   // predecessor rows precomputed per row
 
@@ -88,6 +89,31 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
          (static_cast<uint64_t>(flags) << 48);
 }
 
+// LDS block state, 19 KB total = 8 blocks/CU — exactly the 2048-window
+// batch residency. The Kahn scratch (work + FIFO queue/topological order)
+// aliases the DP row ring: the ring is only read within one layer's DP
+// (every predecessor row is rewritten before use) and the sort runs after
+// the DP, so they never overlap in time. letters/in_cnt/out_cnt/first_out
+// are the canonical copies (global slabs for these are no longer written);
+// full edge lists, weights, rings and rank stay in global slabs.
+struct Shared {
+  union {
+    int16_t ring[kRing][kMaxW];  // DP rows (slot = row % kRing)
+    struct {
+      uint16_t work[kMaxN];   // Kahn in-degree scratch
+      uint16_t queue[kMaxN];  // Kahn FIFO == topological order
+    } kahn;
+  } u;
+  uint16_t first_out[kMaxN];
+  uint8_t letters[kMaxN];
+  uint8_t in_cnt[kMaxN];
+  uint8_t out_cnt[kMaxN];
+  uint8_t seq[kMaxW];
+};
+
+__device__ inline uint16_t out_edge_of(const struct WindowCtx& c, const Shared& s,
+                                       uint32_t node, uint32_t e);
+
 struct WindowCtx {
   uint8_t* letters;
   uint8_t* in_cnt;
@@ -127,11 +153,16 @@ struct WindowCtx {
 
 // ---------- serial (lane 0) graph helpers ----------
 
-__device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t w) {
-  uint32_t n_out = c.out_cnt[a];
+__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
+                                       uint32_t e) {
+  return (e == 0) ? s.first_out[node] : c.out_edges[node * c.ME + e];
+}
+
+__device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
+  uint32_t n_out = s.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
-    if (c.out_edges[a * c.ME + e] == b) {
-      uint32_t n_in = c.in_cnt[b];
+    if (out_edge_of(c, s, a, e) == b) {
+      uint32_t n_in = s.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
         if (c.in_edges[b * c.ME + f] == a) {
           c.in_weights[b * c.ME + f] += w;
@@ -141,28 +172,31 @@ __device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t 
       return true;  // unreachable for a consistent graph
     }
   }
-  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
+  if (n_out >= c.ME || s.in_cnt[b] >= c.ME) {
     c.status = kPoaEdgeOverflow;
     return false;
   }
   c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
-  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
-  uint32_t n_in = c.in_cnt[b];
+  if (n_out == 0) {
+    s.first_out[a] = static_cast<uint16_t>(b);
+  }
+  s.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = s.in_cnt[b];
   c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
   c.in_weights[b * c.ME + n_in] = w;
-  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  s.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
-  if (c.num_nodes >= c.MN) {
+__device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
+  if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
     c.status = kPoaNodeOverflow;
     return -1;
   }
   uint32_t id = c.num_nodes++;
-  c.letters[id] = letter;
-  c.in_cnt[id] = 0;
-  c.out_cnt[id] = 0;
+  s.letters[id] = letter;
+  s.in_cnt[id] = 0;
+  s.out_cnt[id] = 0;
   c.ring_cnt[id] = 0;
   c.nseq[id] = 0;
   return static_cast<int32_t>(id);
@@ -170,8 +204,8 @@ __device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
 
 // Threads the traceback path (stored reversed in aln_*) into the graph.
 // Mirrors Graph::add_alignment (src/align/poa.cpp).
-__device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t* wts,
-                                uint32_t len, int32_t aln_len) {
+__device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
+                                const uint8_t* wts, uint32_t len, int32_t aln_len) {
   // first/last aligned sequence positions
   int32_t first_pos = -1, last_pos = -1;
   for (int32_t k = aln_len - 1; k >= 0; --k) {  // reversed storage -> forward walk
@@ -188,7 +222,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
   int32_t last_counted = -1;
 
   auto link = [&](int32_t a, int32_t b, int32_t w) {
-    if (!add_edge_d(c, a, b, w)) {
+    if (!add_edge_d(c, s, a, b, w)) {
       return;
     }
     if (a != last_counted) {
@@ -205,7 +239,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
 
   // head chain: seq[0 .. first_pos)
   for (int32_t p = 0; p < first_pos; ++p) {
-    int32_t id = add_node_d(c, seq[p]);
+    int32_t id = add_node_d(c, s, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -224,22 +258,22 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
     int32_t node = c.aln_nodes[k];
     int32_t new_id;
     if (node == -1) {
-      new_id = add_node_d(c, letter);
+      new_id = add_node_d(c, s, letter);
       if (new_id < 0) return;
-    } else if (c.letters[node] == letter) {
+    } else if (s.letters[node] == letter) {
       new_id = node;
     } else {
       new_id = -1;
       uint32_t nr = c.ring_cnt[node];
       for (uint32_t r = 0; r < nr; ++r) {
         uint16_t aid = c.ring[node * c.MR + r];
-        if (c.letters[aid] == letter) {
+        if (s.letters[aid] == letter) {
           new_id = aid;
           break;
         }
       }
       if (new_id == -1) {
-        new_id = add_node_d(c, letter);
+        new_id = add_node_d(c, s, letter);
         if (new_id < 0) return;
         // join the ring: new node linked with node and all its partners
         if (nr >= c.MR) {
@@ -273,7 +307,7 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
 
   // tail chain: seq[last_pos+1 .. len)
   for (int32_t p = (last_pos == -1 ? len : last_pos + 1); p < static_cast<int32_t>(len); ++p) {
-    int32_t id = add_node_d(c, seq[p]);
+    int32_t id = add_node_d(c, s, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -285,46 +319,49 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq, const uint8_t*
   ++c.seqs_in_graph;
 }
 
-// Kahn topological sort (FIFO, deterministic). Rebuilds sorted/rank.
-__device__ void topo_sort_d(WindowCtx& c) {
+// Kahn topological sort (FIFO, deterministic) over the LDS mirrors. The
+// FIFO queue IS the final topological order (s.u.kahn.queue); rank goes to
+// the global slab (read lane-parallel by build_row_desc).
+__device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
-    c.work[i] = c.in_cnt[i];
+    s.u.kahn.work[i] = s.in_cnt[i];
   }
   uint32_t qhead = 0, qtail = 0;
   for (uint32_t i = 0; i < n; ++i) {
-    if (c.work[i] == 0) {
-      c.sorted[qtail++] = static_cast<uint16_t>(i);
+    if (s.u.kahn.work[i] == 0) {
+      s.u.kahn.queue[qtail++] = static_cast<uint16_t>(i);
     }
   }
   while (qhead < qtail) {
-    uint16_t u = c.sorted[qhead++];
-    uint32_t nout = c.out_cnt[u];
+    uint16_t u = s.u.kahn.queue[qhead++];
+    uint32_t nout = s.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t v = c.out_edges[u * c.ME + e];
-      if (--c.work[v] == 0) {
-        c.sorted[qtail++] = v;
+      uint16_t v = out_edge_of(c, s, u, e);
+      if (--s.u.kahn.work[v] == 0) {
+        s.u.kahn.queue[qtail++] = v;
       }
     }
   }
   for (uint32_t r = 0; r < qtail; ++r) {
-    c.rank[c.sorted[r]] = static_cast<uint16_t>(r);
+    c.rank[s.u.kahn.queue[r]] = static_cast<uint16_t>(r);
   }
 }
 
 // Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
 // Returns consensus length written into out/cov (forward order), or -1.
-__device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32_t max_out) {
+__device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* cov,
+                               uint32_t max_out) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
     c.hb_score[i] = -1;
     c.hb_pred[i] = -1;
   }
 
-  uint32_t max_id = c.sorted[0];
+  uint32_t max_id = s.u.kahn.queue[0];
   for (uint32_t r = 0; r < n; ++r) {
-    uint16_t nid = c.sorted[r];
-    uint32_t nin = c.in_cnt[nid];
+    uint16_t nid = s.u.kahn.queue[r];
+    uint32_t nin = s.in_cnt[nid];
     for (uint32_t e = 0; e < nin; ++e) {
       uint16_t p = c.in_edges[nid * c.ME + e];
       int64_t w = c.in_weights[nid * c.ME + e];
@@ -348,7 +385,7 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
   // otherwise spin forever — those windows fail over to the CPU instead)
   uint32_t guard = 0;
   uint32_t prev_rank = 0;
-  while (c.out_cnt[max_id] != 0) {
+  while (s.out_cnt[max_id] != 0) {
     if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
       c.status = kPoaConsensusOverflow;
       return -1;
@@ -356,10 +393,10 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
     prev_rank = c.rank[max_id];
     uint32_t rank0 = c.rank[max_id];
     // invalidate alternative branches
-    uint32_t nout = c.out_cnt[max_id];
+    uint32_t nout = s.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t endn = c.out_edges[max_id * c.ME + e];
-      uint32_t nin = c.in_cnt[endn];
+      uint16_t endn = out_edge_of(c, s, max_id, e);
+      uint32_t nin = s.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
         uint16_t o = c.in_edges[endn * c.ME + f];
         if (o != max_id) {
@@ -370,10 +407,10 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
     int64_t best = 0;
     uint32_t best_id = 0;
     for (uint32_t r = rank0 + 1; r < n; ++r) {
-      uint16_t nid = c.sorted[r];
+      uint16_t nid = s.u.kahn.queue[r];
       c.hb_score[nid] = -1;
       c.hb_pred[nid] = -1;
-      uint32_t nin = c.in_cnt[nid];
+      uint32_t nin = s.in_cnt[nid];
       for (uint32_t e = 0; e < nin; ++e) {
         uint16_t p = c.in_edges[nid * c.ME + e];
         if (c.hb_score[p] == -1) {
@@ -412,7 +449,7 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
   idw = static_cast<int32_t>(max_id);
   for (int32_t k = path_len - 1; k >= 0; --k) {
     uint32_t node = static_cast<uint32_t>(idw);
-    out[k] = c.letters[node];
+    out[k] = s.letters[node];
     uint32_t covv = c.nseq[node];
     uint32_t nr = c.ring_cnt[node];
     for (uint32_t r = 0; r < nr; ++r) {
@@ -426,25 +463,25 @@ __device__ int32_t consensus_d(WindowCtx& c, uint8_t* out, uint16_t* cov, uint32
 
 // Lane-parallel: rebuild the packed per-rank row descriptors from the graph
 // arrays (called after backbone init and after every topo sort).
-__device__ void build_row_desc(WindowCtx& c, int lane) {
+__device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
   const uint32_t n = c.num_nodes;
   for (uint32_t node = lane; node < n; node += kLanes) {
     const uint32_t r = c.rank[node];
-    const uint8_t nin = c.in_cnt[node];
+    const uint8_t nin = s.in_cnt[node];
     uint16_t pred_row = 0;
     if (nin > 0) {
       pred_row = static_cast<uint16_t>(c.rank[c.in_edges[node * c.ME]] + 1);
     }
-    const uint32_t nout = c.out_cnt[node];
+    const uint32_t nout = s.out_cnt[node];
     uint8_t flags = (nout == 0) ? kRdEnd : 0;
     for (uint32_t e = 0; e < nout; ++e) {
-      const uint32_t sr = c.rank[c.out_edges[node * c.ME + e]];
+      const uint32_t sr = c.rank[out_edge_of(c, s, node, e)];
       if (sr - r >= kRing) {
         flags |= kRdStore;
         break;
       }
     }
-    c.row_desc[r] = pack_rd(c.letters[node], nin, static_cast<uint16_t>(node), pred_row, flags);
+    c.row_desc[r] = pack_rd(s.letters[node], nin, static_cast<uint16_t>(node), pred_row, flags);
   }
 }
 
@@ -461,8 +498,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const PoaLimits L = a.limits;
   const uint32_t slab = desc.scratch_idx;
 
-  __shared__ int16_t lds_ring[kRing][kMaxW];
-  __shared__ uint8_t lds_seq[kMaxW];
+  __shared__ Shared s;
 
   WindowCtx c;
   c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
@@ -508,33 +544,36 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.m = a.match;
   c.x = a.mismatch;
   c.g = a.gap;
-  c.status = (L.matrix_width <= kMaxW) ? kPoaOk : kPoaNodeOverflow;
+  c.status = (L.matrix_width <= kMaxW && L.max_nodes <= kMaxN) ? kPoaOk : kPoaNodeOverflow;
 
   // ---- init graph from the backbone (layer 0), lane-parallel ----
   const uint32_t bb_len = c.ends[0];
   const uint8_t* bb_seq = c.seq_base;
   const uint8_t* bb_wts = c.weight_base;
   for (uint32_t i = lane; i < bb_len; i += kLanes) {
-    c.letters[i] = bb_seq[i];
+    s.letters[i] = bb_seq[i];
     c.ring_cnt[i] = 0;
     c.nseq[i] = bb_len >= 2 ? 1 : 0;
-    c.sorted[i] = static_cast<uint16_t>(i);
+    // queue holds the trivial order for the no-aligned-layer case (it is
+    // clobbered by the DP ring and rebuilt by every topo sort afterwards)
+    s.u.kahn.queue[i] = static_cast<uint16_t>(i);
     c.rank[i] = static_cast<uint16_t>(i);
     uint8_t nin = 0;
     if (i == 0) {
-      c.in_cnt[i] = 0;
+      s.in_cnt[i] = 0;
     } else {
       nin = 1;
-      c.in_cnt[i] = 1;
+      s.in_cnt[i] = 1;
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
     uint8_t flags = 0;
     if (i + 1 < bb_len) {
-      c.out_cnt[i] = 1;
+      s.out_cnt[i] = 1;
+      s.first_out[i] = static_cast<uint16_t>(i + 1);
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
-      c.out_cnt[i] = 0;
+      s.out_cnt[i] = 0;
       flags = kRdEnd;
     }
     c.row_desc[i] = pack_rd(bb_seq[i], nin, static_cast<uint16_t>(i),
@@ -555,7 +594,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     }
 
     for (uint32_t j = lane; j < len; j += kLanes) {
-      lds_seq[j] = seq[j];
+      s.seq[j] = seq[j];
     }
     __syncthreads();
 
@@ -583,7 +622,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       const bool store_row = (flags & kRdStore) != 0;
       int16_t* Hrow = c.matrix + static_cast<size_t>(r + 1) * c.MW;
       uint8_t* Mrow = c.moves + static_cast<size_t>(r + 1) * c.MW;
-      int16_t* ring_row = lds_ring[(r + 1) % kRing];
+      int16_t* ring_row = s.u.ring[(r + 1) % kRing];
 
       // predecessor rows (e < kMaxPre precomputed; beyond that re-derived
       // in the chunk loop — nodes with >8 in-edges are rare)
@@ -600,7 +639,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           return static_cast<int32_t>(col) * c.g;  // arithmetic row 0
         }
         if (r + 1 - p < kRing) {
-          return lds_ring[p % kRing][col];
+          return s.u.ring[p % kRing][col];
         }
         return c.matrix[static_cast<size_t>(p) * c.MW + col];
       };
@@ -644,7 +683,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         uint32_t ve = 0;
         uint8_t vtype = kMvInvalid;
         if (j < width) {
-          const int32_t sub = (lds_seq[j - 1] == letter) ? c.m : c.x;
+          const int32_t sub = (s.seq[j - 1] == letter) ? c.m : c.x;
           if (nin == 0) {
             // pred is the arithmetic row 0
             const int32_t diag = static_cast<int32_t>(j - 1) * c.g + sub;
@@ -768,11 +807,11 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
 
       t_tb += lap();
       if (c.status == kPoaOk) {
-        add_alignment_d(c, seq, wts, len, aln_len);
+        add_alignment_d(c, s, seq, wts, len, aln_len);
       }
       t_add += lap();
       if (c.status == kPoaOk) {
-        topo_sort_d(c);
+        topo_sort_d(c, s);
       }
       t_topo += lap();
     }
@@ -787,7 +826,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     // rebuild the packed row descriptors for the grown graph, lane-parallel
     (void)lap();
     if (c.status == kPoaOk) {
-      build_row_desc(c, lane);
+      build_row_desc(c, s, lane);
     }
     __syncthreads();
     t_rd += lap();
@@ -800,7 +839,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     int32_t clen = -1;
     (void)lap();
     if (c.status == kPoaOk) {
-      clen = consensus_d(c, out, cov, L.max_consensus);
+      clen = consensus_d(c, s, out, cov, L.max_consensus);
     }
     t_cons = lap();
     a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
