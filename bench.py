@@ -47,7 +47,8 @@ def main():
                     help="genome Mbp per GPU shard (default 12.5 = C.elegans/8)")
     ap.add_argument("--coverage", type=int, default=30)
     ap.add_argument("--window", type=int, default=500)
-    ap.add_argument("--threads", type=int, default=max(4, (os.cpu_count() or 8) // 8))
+    ap.add_argument("--threads", type=int, default=None,
+                    help="CPU threads per rank (default: ncpu / (2*world))")
     ap.add_argument("--poa-batches", type=int, default=8)
     ap.add_argument("--aligner-batches", type=int, default=4)
     ap.add_argument("--cpu", action="store_true", help="force CPU path (debug)")
@@ -56,6 +57,8 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     n_gpus = max(args.gpus, world)
+    if args.threads is None:
+        args.threads = max(8, (os.cpu_count() or 16) // (2 * world))
 
     have_gpu = torch.cuda.is_available() and not args.cpu
     # Honest default: the BASELINE config is 12.5 Mbp/GPU (100 Mbp at 8 GPUs).
