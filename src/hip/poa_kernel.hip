@@ -89,13 +89,15 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
          (static_cast<uint64_t>(flags) << 48);
 }
 
-// LDS block state, 19 KB total = 8 blocks/CU — exactly the 2048-window
-// batch residency. The Kahn scratch (work + FIFO queue/topological order)
-// aliases the DP row ring: the ring is only read within one layer's DP
-// (every predecessor row is rewritten before use) and the sort runs after
-// the DP, so they never overlap in time. letters/in_cnt/out_cnt/first_out
-// are the canonical copies (global slabs for these are no longer written);
-// full edge lists, weights, rings and rank stay in global slabs.
+// LDS block state, 9.2 KB = 17 blocks/CU. Cross-batch window co-residency
+// is the dominant throughput lever (a 19 KB variant with full graph-array
+// mirrors made each window ~10% faster but halved residency and lost 20%
+// end to end), so only two things live here: the DP row ring, and — for
+// free, aliased into the same union because the phases never overlap — the
+// Kahn scratch + FIFO queue, which kills the store-to-load round trip the
+// serial sort otherwise does against HBM for every node. The queue doubles
+// as the topological order read by the consensus phase. Everything else
+// stays in the global slabs, hidden by co-residency.
 struct Shared {
   union {
     int16_t ring[kRing][kMaxW];  // DP rows (slot = row % kRing)
@@ -104,10 +106,6 @@ struct Shared {
       uint16_t queue[kMaxN];  // Kahn FIFO == topological order
     } kahn;
   } u;
-  uint16_t first_out[kMaxN];
-  uint8_t letters[kMaxN];
-  uint8_t in_cnt[kMaxN];
-  uint8_t out_cnt[kMaxN];
   uint8_t seq[kMaxW];
 };
 
@@ -153,16 +151,16 @@ struct WindowCtx {
 
 // ---------- serial (lane 0) graph helpers ----------
 
-__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared& s, uint32_t node,
+__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared&, uint32_t node,
                                        uint32_t e) {
-  return (e == 0) ? s.first_out[node] : c.out_edges[node * c.ME + e];
+  return c.out_edges[node * c.ME + e];
 }
 
 __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
-  uint32_t n_out = s.out_cnt[a];
+  uint32_t n_out = c.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
     if (out_edge_of(c, s, a, e) == b) {
-      uint32_t n_in = s.in_cnt[b];
+      uint32_t n_in = c.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
         if (c.in_edges[b * c.ME + f] == a) {
           c.in_weights[b * c.ME + f] += w;
@@ -172,31 +170,28 @@ __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t 
       return true;  // unreachable for a consistent graph
     }
   }
-  if (n_out >= c.ME || s.in_cnt[b] >= c.ME) {
+  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
     c.status = kPoaEdgeOverflow;
     return false;
   }
   c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
-  if (n_out == 0) {
-    s.first_out[a] = static_cast<uint16_t>(b);
-  }
-  s.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
-  uint32_t n_in = s.in_cnt[b];
+  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = c.in_cnt[b];
   c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
   c.in_weights[b * c.ME + n_in] = w;
-  s.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
+__device__ inline int32_t add_node_d(WindowCtx& c, Shared&, uint8_t letter) {
   if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
     c.status = kPoaNodeOverflow;
     return -1;
   }
   uint32_t id = c.num_nodes++;
-  s.letters[id] = letter;
-  s.in_cnt[id] = 0;
-  s.out_cnt[id] = 0;
+  c.letters[id] = letter;
+  c.in_cnt[id] = 0;
+  c.out_cnt[id] = 0;
   c.ring_cnt[id] = 0;
   c.nseq[id] = 0;
   return static_cast<int32_t>(id);
@@ -260,14 +255,14 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
     if (node == -1) {
       new_id = add_node_d(c, s, letter);
       if (new_id < 0) return;
-    } else if (s.letters[node] == letter) {
+    } else if (c.letters[node] == letter) {
       new_id = node;
     } else {
       new_id = -1;
       uint32_t nr = c.ring_cnt[node];
       for (uint32_t r = 0; r < nr; ++r) {
         uint16_t aid = c.ring[node * c.MR + r];
-        if (s.letters[aid] == letter) {
+        if (c.letters[aid] == letter) {
           new_id = aid;
           break;
         }
@@ -325,7 +320,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
 __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
-    s.u.kahn.work[i] = s.in_cnt[i];
+    s.u.kahn.work[i] = c.in_cnt[i];
   }
   uint32_t qhead = 0, qtail = 0;
   for (uint32_t i = 0; i < n; ++i) {
@@ -335,7 +330,7 @@ __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   }
   while (qhead < qtail) {
     uint16_t u = s.u.kahn.queue[qhead++];
-    uint32_t nout = s.out_cnt[u];
+    uint32_t nout = c.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
       uint16_t v = out_edge_of(c, s, u, e);
       if (--s.u.kahn.work[v] == 0) {
@@ -361,7 +356,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   uint32_t max_id = s.u.kahn.queue[0];
   for (uint32_t r = 0; r < n; ++r) {
     uint16_t nid = s.u.kahn.queue[r];
-    uint32_t nin = s.in_cnt[nid];
+    uint32_t nin = c.in_cnt[nid];
     for (uint32_t e = 0; e < nin; ++e) {
       uint16_t p = c.in_edges[nid * c.ME + e];
       int64_t w = c.in_weights[nid * c.ME + e];
@@ -385,7 +380,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   // otherwise spin forever — those windows fail over to the CPU instead)
   uint32_t guard = 0;
   uint32_t prev_rank = 0;
-  while (s.out_cnt[max_id] != 0) {
+  while (c.out_cnt[max_id] != 0) {
     if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
       c.status = kPoaConsensusOverflow;
       return -1;
@@ -393,10 +388,10 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
     prev_rank = c.rank[max_id];
     uint32_t rank0 = c.rank[max_id];
     // invalidate alternative branches
-    uint32_t nout = s.out_cnt[max_id];
+    uint32_t nout = c.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
       uint16_t endn = out_edge_of(c, s, max_id, e);
-      uint32_t nin = s.in_cnt[endn];
+      uint32_t nin = c.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
         uint16_t o = c.in_edges[endn * c.ME + f];
         if (o != max_id) {
@@ -410,7 +405,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
       uint16_t nid = s.u.kahn.queue[r];
       c.hb_score[nid] = -1;
       c.hb_pred[nid] = -1;
-      uint32_t nin = s.in_cnt[nid];
+      uint32_t nin = c.in_cnt[nid];
       for (uint32_t e = 0; e < nin; ++e) {
         uint16_t p = c.in_edges[nid * c.ME + e];
         if (c.hb_score[p] == -1) {
@@ -449,7 +444,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   idw = static_cast<int32_t>(max_id);
   for (int32_t k = path_len - 1; k >= 0; --k) {
     uint32_t node = static_cast<uint32_t>(idw);
-    out[k] = s.letters[node];
+    out[k] = c.letters[node];
     uint32_t covv = c.nseq[node];
     uint32_t nr = c.ring_cnt[node];
     for (uint32_t r = 0; r < nr; ++r) {
@@ -467,12 +462,12 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
   const uint32_t n = c.num_nodes;
   for (uint32_t node = lane; node < n; node += kLanes) {
     const uint32_t r = c.rank[node];
-    const uint8_t nin = s.in_cnt[node];
+    const uint8_t nin = c.in_cnt[node];
     uint16_t pred_row = 0;
     if (nin > 0) {
       pred_row = static_cast<uint16_t>(c.rank[c.in_edges[node * c.ME]] + 1);
     }
-    const uint32_t nout = s.out_cnt[node];
+    const uint32_t nout = c.out_cnt[node];
     uint8_t flags = (nout == 0) ? kRdEnd : 0;
     for (uint32_t e = 0; e < nout; ++e) {
       const uint32_t sr = c.rank[out_edge_of(c, s, node, e)];
@@ -481,7 +476,7 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
         break;
       }
     }
-    c.row_desc[r] = pack_rd(s.letters[node], nin, static_cast<uint16_t>(node), pred_row, flags);
+    c.row_desc[r] = pack_rd(c.letters[node], nin, static_cast<uint16_t>(node), pred_row, flags);
   }
 }
 
@@ -551,7 +546,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const uint8_t* bb_seq = c.seq_base;
   const uint8_t* bb_wts = c.weight_base;
   for (uint32_t i = lane; i < bb_len; i += kLanes) {
-    s.letters[i] = bb_seq[i];
+    c.letters[i] = bb_seq[i];
     c.ring_cnt[i] = 0;
     c.nseq[i] = bb_len >= 2 ? 1 : 0;
     // queue holds the trivial order for the no-aligned-layer case (it is
@@ -560,20 +555,19 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     c.rank[i] = static_cast<uint16_t>(i);
     uint8_t nin = 0;
     if (i == 0) {
-      s.in_cnt[i] = 0;
+      c.in_cnt[i] = 0;
     } else {
       nin = 1;
-      s.in_cnt[i] = 1;
+      c.in_cnt[i] = 1;
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
     uint8_t flags = 0;
     if (i + 1 < bb_len) {
-      s.out_cnt[i] = 1;
-      s.first_out[i] = static_cast<uint16_t>(i + 1);
+      c.out_cnt[i] = 1;
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
-      s.out_cnt[i] = 0;
+      c.out_cnt[i] = 0;
       flags = kRdEnd;
     }
     c.row_desc[i] = pack_rd(bb_seq[i], nin, static_cast<uint16_t>(i),
