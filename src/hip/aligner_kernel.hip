@@ -115,10 +115,38 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
   }
 
   // ---- column loop (uniform bound; finished lanes coast) ----
+  // Eq words for the next column are prefetched during the current column's
+  // recurrence: the gather (64 lanes x scattered 8 B) is the only global
+  // read in the loop and would otherwise sit on the critical path.
+  uint64_t EqN[K];
+  int32_t btopN = 0;
+  if (m > 0) {
+    const uint32_t c1 = base_code(t[0]);
+    btopN = btop_of<K>(1, step, nbt);
+#pragma unroll
+    for (int b = 0; b < K; ++b) {
+      EqN[b] =
+          (c1 < 4) ? peq[(static_cast<uint64_t>(btopN + b) * 4 + c1) * kLanes + lane] : 0ull;
+    }
+  }
   for (int32_t j = 1; j <= static_cast<int32_t>(wd.mmax); ++j) {
     if (j <= m) {
-      const uint32_t c = base_code(t[j - 1]);
-      const int32_t btop_new = btop_of<K>(j, step, nbt);
+      uint64_t EqC[K];
+#pragma unroll
+      for (int b = 0; b < K; ++b) {
+        EqC[b] = EqN[b];
+      }
+      const int32_t btop_new = btopN;
+      if (j < m) {  // issue next column's gather before the compute chain
+        const uint32_t cn = base_code(t[j]);
+        btopN = btop_of<K>(j + 1, step, nbt);
+#pragma unroll
+        for (int b = 0; b < K; ++b) {
+          EqN[b] = (cn < 4)
+                       ? peq[(static_cast<uint64_t>(btopN + b) * 4 + cn) * kLanes + lane]
+                       : 0ull;
+        }
+      }
       while (btop < btop_new) {
         // band slides down one block: drop top, append pessimistic bottom
 #pragma unroll
@@ -136,8 +164,7 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
       int32_t hin = 1;  // top boundary: D(top-1, j) - D(top-1, j-1) = +1
 #pragma unroll
       for (int b = 0; b < K; ++b) {
-        uint64_t Eq =
-            (c < 4) ? peq[(static_cast<uint64_t>(btop + b) * 4 + c) * kLanes + lane] : 0ull;
+        uint64_t Eq = EqC[b];
         const uint64_t hin_neg = (hin < 0) ? 1ull : 0ull;
         const uint64_t hin_pos = (hin > 0) ? 1ull : 0ull;
         const uint64_t Xv = Eq | Mv[b];

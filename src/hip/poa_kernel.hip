@@ -106,8 +106,22 @@ struct Shared {
       uint16_t queue[kMaxN];  // Kahn FIFO == topological order
     } kahn;
   } u;
+  // per-layer match bitvectors: bit l of match[c][k] says layer base
+  // (1 + k*64 + l) equals code c — turns the per-cell seq comparison into
+  // one register bit test instead of an LDS byte read
+  uint64_t match[4][kMaxW / 64];
   uint8_t seq[kMaxW];
 };
+
+__device__ inline int32_t poa_code(uint8_t b) {
+  switch (b) {
+    case 'A': return 0;
+    case 'C': return 1;
+    case 'G': return 2;
+    case 'T': return 3;
+    default: return -1;
+  }
+}
 
 __device__ inline uint16_t out_edge_of(const struct WindowCtx& c, const Shared& s,
                                        uint32_t node, uint32_t e);
@@ -590,6 +604,22 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     for (uint32_t j = lane; j < len; j += kLanes) {
       s.seq[j] = seq[j];
     }
+    // build the per-layer match bitvectors (lane-parallel over chunk words)
+    {
+      const uint32_t words = (len + kLanes - 1) / kLanes;
+      for (uint32_t w = lane; w < words * 4; w += kLanes) {
+        const uint32_t k = w >> 2, cc = w & 3;
+        uint64_t bits = 0;
+        const uint32_t base = k * kLanes;
+        const uint32_t lim = min(kLanes, len - base);
+        for (uint32_t l = 0; l < lim; ++l) {
+          if (poa_code(seq[base + l]) == static_cast<int32_t>(cc)) {
+            bits |= 1ull << l;
+          }
+        }
+        s.match[cc][k] = bits;
+      }
+    }
     __syncthreads();
 
     const uint32_t n = c.num_nodes;
@@ -608,6 +638,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         rd_next = c.row_desc[r + 1];  // prefetch: off the critical path
       }
       const uint8_t letter = static_cast<uint8_t>(rd);
+      const int32_t letter_code = poa_code(letter);
       const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
       const uint32_t node = static_cast<uint32_t>((rd >> 16) & 0xffff);
       const uint32_t pred0 = static_cast<uint32_t>((rd >> 32) & 0xffff);
@@ -677,7 +708,11 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         uint32_t ve = 0;
         uint8_t vtype = kMvInvalid;
         if (j < width) {
-          const int32_t sub = (s.seq[j - 1] == letter) ? c.m : c.x;
+          // letter_code < 0 (non-ACGT letter): rare — byte-compare fallback
+          const bool is_match = (letter_code >= 0)
+                                    ? ((s.match[letter_code][k] >> lane) & 1) != 0
+                                    : (s.seq[j - 1] == letter);
+          const int32_t sub = is_match ? c.m : c.x;
           if (nin == 0) {
             // pred is the arithmetic row 0
             const int32_t diag = static_cast<int32_t>(j - 1) * c.g + sub;
