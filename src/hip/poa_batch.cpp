@@ -34,14 +34,13 @@ size_t slab_bytes(const PoaLimits& L) {
 PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch, int8_t gap,
                    bool banded, uint32_t max_depth)
     : device_(device), match_(match), mismatch_(mismatch), gap_(gap), max_depth_(max_depth) {
-  (void)banded;  // v1 always runs the full-width band (reference default mode)
-
   // int16 score guard: worst |score| <= (max_nodes + matrix_width) * max|param|
+  // (banded mode additionally reserves values below -28000 as out-of-band)
   int32_t worst = static_cast<int32_t>(limits_.max_nodes + limits_.matrix_width) *
                   std::max({std::abs(static_cast<int32_t>(match)),
                             std::abs(static_cast<int32_t>(mismatch)),
                             std::abs(static_cast<int32_t>(gap))});
-  if (worst > 32000) {
+  if (worst > (banded ? 27000 : 32000)) {
     fprintf(stderr,
             "[rga::hip::PoaBatch] error: score parameters too large for int16 "
             "GPU scores (|m|,|x|,|g| must keep (%u+%u)*max <= 32000)\n",
@@ -149,6 +148,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.match = match_;
   arena_.mismatch = mismatch_;
   arena_.gap = gap_;
+  arena_.band_width = banded ? 256 : 0;  // reference static band 256
   arena_.limits = limits_;
 }
 

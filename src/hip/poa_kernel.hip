@@ -75,6 +75,22 @@ __device__ inline int32_t wave_scan_max(int32_t v, int /*lane*/) {
   return v;
 }
 
+// chunk range [klo, khi] of the static band for DP row R (1-based):
+// center column follows the rank diagonal via a 16.16 slope; the band is
+// rounded out to whole 64-column chunks (>= bw coverage). Full-width mode
+// (bw == 0) is the caller's fast path and never calls this.
+__device__ inline void band_chunks(uint32_t R, uint32_t slope16, uint32_t bw, uint32_t chunks,
+                                   uint32_t* klo, uint32_t* khi) {
+  const uint32_t center = (static_cast<uint64_t>(R) * slope16) >> 16;
+  const uint32_t jlo = center > bw / 2 ? center - bw / 2 : 0;
+  uint32_t lo = (jlo == 0) ? 0 : (jlo - 1) / 64;
+  if (lo > chunks - 1) lo = chunks - 1;
+  uint32_t hi = lo + bw / 64;
+  if (hi > chunks - 1) hi = chunks - 1;
+  *klo = lo;
+  *khi = hi;
+}
+
 // packed row descriptor (one per topological rank); flags: bit0 = end node
 // (no out-edges), bit1 = some successor is beyond the LDS ring, so this
 // row's scores must go to global memory (matrix rows are otherwise
@@ -156,6 +172,7 @@ struct WindowCtx {
   uint32_t MR;  // max ring
   uint32_t MW;  // matrix width
   uint32_t MN;  // max nodes
+  uint32_t bw;  // 0 = full width; else static band (columns) around diagonal
   int32_t m, x, g;
 
   uint32_t num_nodes;
@@ -550,6 +567,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.MR = L.max_ring;
   c.MW = L.matrix_width;
   c.MN = L.max_nodes;
+  c.bw = a.band_width;
   c.m = a.match;
   c.x = a.mismatch;
   c.g = a.gap;
@@ -625,6 +643,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     const uint32_t n = c.num_nodes;
     const uint32_t width = len + 1;
     const uint32_t chunks = (len + kLanes - 1) / kLanes;
+    // banded mode (-b): static band around the rank diagonal
+    const bool banded = (c.bw != 0) && (c.bw < len);
+    const uint32_t slope16 =
+        banded ? static_cast<uint32_t>((static_cast<uint64_t>(len) << 16) / n) : 0;
 
     int32_t best_score = kNegInf;
     uint32_t best_row = 0;
@@ -658,10 +680,30 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         pred_rows[e] = c.rank[c.in_edges[node * c.ME + e]] + 1;
       }
 
-      // fetch a predecessor row value: LDS ring if close, global otherwise
+      uint32_t row_klo = 0, row_khi = chunks - 1;
+      if (banded) {
+        band_chunks(r + 1, slope16, c.bw, chunks, &row_klo, &row_khi);
+      }
+
+      // fetch a predecessor row value: LDS ring if close, global otherwise;
+      // in banded mode, columns outside the predecessor's band are -inf
       auto pred_val = [&](uint32_t p, uint32_t col) -> int32_t {
         if (p == 0) {
           return static_cast<int32_t>(col) * c.g;  // arithmetic row 0
+        }
+        if (banded && col != 0) {
+          uint32_t pklo, pkhi;
+          band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+          const uint32_t kc = (col - 1) / 64;
+          if (kc < pklo || kc > pkhi) {
+            return kNegInf;
+          }
+        } else if (banded && col == 0) {
+          uint32_t pklo, pkhi;
+          band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+          if (pklo != 0) {
+            return kNegInf;
+          }
         }
         if (r + 1 - p < kRing) {
           return s.u.ring[p % kRing][col];
@@ -670,9 +712,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       };
 
       // first column (j = 0): max over preds of Hp[0] + gap
-      int32_t h0;
+      // (banded: only when this row's band includes column 0)
+      int32_t h0 = kNegInf;
       uint32_t e0 = 0;
-      {
+      if (row_klo == 0) {
         int32_t best0 = kNegInf;
         if (nin == 0) {
           best0 = 0;
@@ -702,7 +745,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       int32_t carry_u = h0;
       int32_t last_col_val = kNegInf;
 
-      for (uint32_t k = 0; k < chunks; ++k) {
+      for (uint32_t k = row_klo; k <= row_khi; ++k) {
         const uint32_t j = 1 + k * kLanes + lane;  // column this lane owns
         int32_t v = kNegInf;
         uint32_t ve = 0;
@@ -754,10 +797,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         u = max(u, carry_u);
         const int32_t h = u + static_cast<int32_t>(j) * c.g;
         if (j < width) {
+          // int16 clamp: banded -inf propagation must stay very negative
+          const int32_t h16 = h < -28000 ? -28000 : h;
           if (store_row) {
-            Hrow[j] = static_cast<int16_t>(h);
+            Hrow[j] = static_cast<int16_t>(h16);
           }
-          ring_row[j] = static_cast<int16_t>(h);
+          ring_row[j] = static_cast<int16_t>(h16);
           // left move only when the scan strictly beat this cell's v
           Mrow[j] = (h == v) ? static_cast<uint8_t>(vtype | (ve << 2)) : kMvLeft;
           if (j == len) {

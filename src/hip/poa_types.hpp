@@ -89,6 +89,8 @@ struct PoaDeviceArena {
   int32_t* status;          // [1] per window
 
   int8_t match, mismatch, gap;
+  uint32_t band_width;  // 0 = full-width DP; else static band (reference -b:
+                        // BatchConfig band 256, src/cuda/cudabatch.cpp:56-59)
   PoaLimits limits;
 };
 

@@ -103,6 +103,22 @@ def test_gpu_matches_cpu_closely(racon, sample):
     assert ed < 0.005 * len(cpu[0][1]), ed
 
 
+def test_gpu_banded_poa(racon, sample, fasta_reader):
+    """-b static-band POA: approximation must stay within a small divergence
+    of the full-width result (reference capability: BatchConfig static_band,
+    src/cuda/cudabatch.cpp:56-59)."""
+    truth = list(fasta_reader(sample["reference"]).values())[0]
+    full = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                        threads=4, poa_batches=1)
+    band = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                        threads=4, poa_batches=1, banded_poa=True)
+    assert len(band) == 1
+    ed_full = racon.edit_distance(full[0][1], truth)
+    ed_band = racon.edit_distance(band[0][1], truth)
+    # banded consensus quality within 2x of full-width error (both tiny)
+    assert ed_band <= max(ed_full * 2, ed_full + 20), (ed_full, ed_band)
+
+
 def test_gpu_deterministic(racon, sample):
     a = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
                      threads=4, poa_batches=1)
