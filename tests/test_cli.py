@@ -74,3 +74,28 @@ def test_cli_polishes_sample(racon_cli, sample):
     # reference src/polisher.cpp:522-525)
     assert "LN:i:" in lines[0] and "RC:i:" in lines[0] and "XC:f:" in lines[0]
     assert len("".join(lines[1:])) > 10000
+
+
+def test_cudapoa_batches_optional_arg(racon_cli, sample):
+    """-c takes an optional argument defaulting to 1 (reference quirk,
+    src/main.cpp:114-126). On a GPU-less host selecting the GPU pipeline
+    fails loudly - which proves the flag parsed and routed."""
+    import _racon
+    if _racon.device_count() > 0:
+        return  # covered by GPU tests on GPU hosts
+    for args in (["-c", "2"], ["--cudaaligner-batches", "2"]):
+        out = run_racon(racon_cli, args + [sample["reads"], sample["overlaps"],
+                                           sample["layout"]])
+        assert out.returncode != 0
+        assert "no HIP devices" in out.stderr
+    # -c followed by another option still defaults to 1 (not consumed)
+    out = run_racon(racon_cli, ["-c", "-t", "2", sample["reads"], sample["overlaps"],
+                                sample["layout"]])
+    assert out.returncode != 0
+    assert "no HIP devices" in out.stderr
+    # reference quirk preserved exactly (main.cpp:114-126): a bare -c CONSUMES
+    # the next non-dash token - here the reads path - so input files go missing
+    out = run_racon(racon_cli, ["-c", sample["reads"], sample["overlaps"],
+                                sample["layout"]])
+    assert out.returncode != 0
+    assert "missing input file" in out.stderr
