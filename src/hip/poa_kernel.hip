@@ -26,6 +26,8 @@
 // GPU goldens; so do we — topological order here is Kahn FIFO, not spoa DFS).
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #include "hip/poa_types.hpp"
 
 namespace rga::hip {
@@ -37,7 +39,8 @@ constexpr int32_t kNegInf = -(1 << 28);
 constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
 constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
-constexpr uint32_t kMaxPre = 8;   // predecessor rows precomputed per row
+constexpr uint32_t kMaxPre = 4;   // predecessor rows precomputed per row
+constexpr uint32_t kWB = 8;       // contiguous columns per lane per DP pass
 
 // move byte encoding
 constexpr uint8_t kMvDiag = 0;
@@ -124,8 +127,9 @@ struct Shared {
   } u;
   // per-layer match bitvectors: bit l of match[c][k] says layer base
   // (1 + k*64 + l) equals code c — turns the per-cell seq comparison into
-  // one register bit test instead of an LDS byte read
-  uint64_t match[4][kMaxW / 64];
+  // one register bit test instead of an LDS byte read. One spare word so
+  // cross-word extraction at the last chunk never reads out of bounds.
+  uint64_t match[4][kMaxW / 64 + 1];
   uint8_t seq[kMaxW];
 };
 
@@ -513,6 +517,8 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
 
 // ---------- the mega-kernel ----------
 
+template <bool TIMED>
+__attribute__((amdgpu_waves_per_eu(5)))
 __launch_bounds__(kLanes)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   const uint32_t win = blockIdx.x;
@@ -548,11 +554,14 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.row_desc = a.row_desc + static_cast<size_t>(slab) * L.max_nodes;
 
   unsigned long long* timing = a.timing + static_cast<size_t>(win) * 8;
-  unsigned long long tick = wall_clock64();
+  unsigned long long tick = TIMED ? wall_clock64() : 0;
   const unsigned long long t_start = tick;
   unsigned long long t_dp = 0, t_tb = 0, t_add = 0, t_topo = 0, t_rd = 0, t_cons = 0;
   unsigned long long layers_done = 0;
-  auto lap = [&]() {
+  auto lap = [&]() -> unsigned long long {
+    if (!TIMED) {
+      return 0;
+    }
     unsigned long long now = wall_clock64();
     unsigned long long d = now - tick;
     tick = now;
@@ -622,17 +631,20 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     for (uint32_t j = lane; j < len; j += kLanes) {
       s.seq[j] = seq[j];
     }
-    // build the per-layer match bitvectors (lane-parallel over chunk words)
+    // build the per-layer match bitvectors (lane-parallel over chunk words;
+    // one spare zero word past the end for cross-word extraction)
     {
-      const uint32_t words = (len + kLanes - 1) / kLanes;
+      const uint32_t words = (len + kLanes - 1) / kLanes + 1;
       for (uint32_t w = lane; w < words * 4; w += kLanes) {
         const uint32_t k = w >> 2, cc = w & 3;
         uint64_t bits = 0;
         const uint32_t base = k * kLanes;
-        const uint32_t lim = min(kLanes, len - base);
-        for (uint32_t l = 0; l < lim; ++l) {
-          if (poa_code(seq[base + l]) == static_cast<int32_t>(cc)) {
-            bits |= 1ull << l;
+        if (base < len) {
+          const uint32_t lim = min(kLanes, len - base);
+          for (uint32_t l = 0; l < lim; ++l) {
+            if (poa_code(seq[base + l]) == static_cast<int32_t>(cc)) {
+              bits |= 1ull << l;
+            }
           }
         }
         s.match[cc][k] = bits;
@@ -741,89 +753,176 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         }
       }
 
-      // carry for the horizontal scan, in u-space: u(j) = H[j] - j*g
-      int32_t carry_u = h0;
+      // ---- lane-blocked columns ----
+      // Lane l owns kWB contiguous columns per pass; one register-local
+      // inclusive scan + one DPP wave scan per pass replaces the previous
+      // chunk-serial carry chain (8 dependent LDS+scan segments per row).
+      const uint32_t j0 = banded ? row_klo * kLanes : 0;
+      const uint32_t jend = banded ? min(len, (row_khi + 1) * kLanes) : len;
+      int32_t carry_u = h0;  // u-space max through column j0 (h0 = -inf when
+                             // the band excludes column 0)
       int32_t last_col_val = kNegInf;
 
-      for (uint32_t k = row_klo; k <= row_khi; ++k) {
-        const uint32_t j = 1 + k * kLanes + lane;  // column this lane owns
-        int32_t v = kNegInf;
-        uint32_t ve = 0;
-        uint8_t vtype = kMvInvalid;
-        if (j < width) {
-          // letter_code < 0 (non-ACGT letter): rare — byte-compare fallback
-          const bool is_match = (letter_code >= 0)
-                                    ? ((s.match[letter_code][k] >> lane) & 1) != 0
-                                    : (s.seq[j - 1] == letter);
-          const int32_t sub = is_match ? c.m : c.x;
-          if (nin == 0) {
-            // pred is the arithmetic row 0
-            const int32_t diag = static_cast<int32_t>(j - 1) * c.g + sub;
-            const int32_t up = static_cast<int32_t>(j) * c.g + c.g;
-            v = max(diag, up);
-            vtype = (diag >= up) ? kMvDiag : kMvUp;
-          } else {
-            int32_t best_diag = kNegInf, best_up = kNegInf;
-            uint32_t e_diag = 0, e_up = 0;
-            for (uint32_t e = 0; e < nin; ++e) {
-              const uint32_t p = (e < kMaxPre) ? pred_rows[e]
-                                               : c.rank[c.in_edges[node * c.ME + e]] + 1;
-              const int32_t hpjm1 = pred_val(p, j - 1);
-              const int32_t hpj = pred_val(p, j);
-              if (hpjm1 + sub > best_diag) {
-                best_diag = hpjm1 + sub;
-                e_diag = e;
-              }
-              if (hpj + c.g > best_up) {
-                best_up = hpj + c.g;
-                e_up = e;
-              }
+      for (uint32_t base = j0; base < jend; base += kLanes * kWB) {
+        const uint32_t cbase = base + lane * kWB;  // own cols: cbase+1..cbase+kWB
+        const uint32_t nown =
+            (cbase < jend) ? min(kWB, jend - cbase) : 0;
+
+        // substitution matches for own columns: bits w of mbits
+        uint64_t mbits = 0;
+        if (letter_code >= 0 && nown > 0) {
+          const uint32_t bit0 = cbase & 63;
+          const uint32_t w0 = cbase >> 6;
+          mbits = s.match[letter_code][w0] >> bit0;
+          if (bit0 != 0) {
+            mbits |= s.match[letter_code][w0 + 1] << (64 - bit0);
+          }
+        } else if (nown > 0) {
+          for (uint32_t w = 0; w < nown; ++w) {
+            if (s.seq[cbase + w] == letter) {
+              mbits |= 1ull << w;
             }
-            // traceback priority: diagonal first, then vertical
-            if (best_diag >= best_up) {
-              v = best_diag;
-              vtype = kMvDiag;
-              ve = e_diag;
+          }
+        }
+
+        int32_t bd[kWB], bu[kWB];  // best diagonal / vertical candidates
+#pragma unroll
+        for (uint32_t w = 0; w < kWB; ++w) {
+          bd[w] = kNegInf;
+          bu[w] = kNegInf;
+        }
+
+        if (nin == 0) {
+#pragma unroll
+          for (uint32_t w = 0; w < kWB; ++w) {
+            if (w < nown) {
+              const int32_t j = static_cast<int32_t>(cbase + 1 + w);
+              const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
+              bd[w] = (j - 1) * c.g + sub;
+              bu[w] = j * c.g + c.g;
+            }
+          }
+        } else {
+          for (uint32_t e = 0; e < nin; ++e) {
+            const uint32_t p = (e < kMaxPre) ? pred_rows[e]
+                                             : c.rank[c.in_edges[node * c.ME + e]] + 1;
+            // gather pred row values pv[w] = H(p, cbase + w), w in 0..kWB
+            int32_t pv[kWB + 1];
+            if (p == 0) {
+#pragma unroll
+              for (uint32_t w = 0; w <= kWB; ++w) {
+                pv[w] = static_cast<int32_t>(cbase + w) * c.g;
+              }
             } else {
-              v = best_up;
-              vtype = kMvUp;
-              ve = e_up;
+              uint32_t plo = 0, phi = len;  // valid column range of pred row
+              bool p_has0 = true;
+              if (banded) {
+                uint32_t pklo, pkhi;
+                band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+                plo = pklo * kLanes;  // valid cols: {0 if pklo==0} + [plo+1..phi]
+                phi = min(len, (pkhi + 1) * kLanes);
+                p_has0 = (pklo == 0);
+              }
+              const int16_t* src = (r + 1 - p < kRing)
+                                       ? s.u.ring[p % kRing]
+                                       : c.matrix + static_cast<size_t>(p) * c.MW;
+#pragma unroll
+              for (uint32_t w = 0; w <= kWB; ++w) {
+                const uint32_t col = cbase + w;
+                const bool ok =
+                    (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
+                pv[w] = ok ? static_cast<int32_t>(src[col]) : kNegInf;
+              }
+            }
+#pragma unroll
+            for (uint32_t w = 0; w < kWB; ++w) {
+              if (w < nown) {
+                const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
+                bd[w] = max(bd[w], pv[w] + sub);
+                bu[w] = max(bu[w], pv[w + 1] + c.g);
+              }
             }
           }
         }
-        // horizontal pass: H[j] = max(v[j], H[j-1] + g) as a max-scan in u-space
-        int32_t u = (j < width) ? v - static_cast<int32_t>(j) * c.g : kNegInf;
-        u = wave_scan_max(u, lane);
-        u = max(u, carry_u);
-        const int32_t h = u + static_cast<int32_t>(j) * c.g;
-        if (j < width) {
-          // int16 clamp: banded -inf propagation must stay very negative
-          const int32_t h16 = h < -28000 ? -28000 : h;
-          if (store_row) {
-            Hrow[j] = static_cast<int16_t>(h16);
-          }
-          ring_row[j] = static_cast<int16_t>(h16);
-          // left move only when the scan strictly beat this cell's v
-          Mrow[j] = (h == v) ? static_cast<uint8_t>(vtype | (ve << 2)) : kMvLeft;
-          if (j == len) {
-            last_col_val = h;
+
+        // local inclusive u-space scan over own columns
+        int32_t us[kWB];
+        int32_t run = kNegInf;
+#pragma unroll
+        for (uint32_t w = 0; w < kWB; ++w) {
+          const int32_t j = static_cast<int32_t>(cbase + 1 + w);
+          const int32_t v = max(bd[w], bu[w]);
+          const int32_t u = (w < nown) ? v - j * c.g : kNegInf;
+          run = max(run, u);
+          us[w] = run;
+        }
+        // wave scan over lane totals -> exclusive prefix for this lane
+        const int32_t incl = wave_scan_max(run, lane);
+        int32_t excl =
+            __builtin_amdgcn_update_dpp(kNegInf, incl, 0x138, 0xf, 0xf, false);  // wave_shr:1
+        excl = max(excl, carry_u);
+        // next pass's carry: wave-uniform max over everything <= this pass
+        carry_u = max(carry_u, __builtin_amdgcn_readlane(incl, kLanes - 1));
+
+        // finalize own columns: h, moves, stores
+        int32_t h_sel = kNegInf;
+#pragma unroll
+        for (uint32_t w = 0; w < kWB; ++w) {
+          if (w < nown) {
+            const uint32_t j = cbase + 1 + w;
+            const int32_t v = max(bd[w], bu[w]);
+            const int32_t h = max(us[w], excl) + static_cast<int32_t>(j) * c.g;
+            uint8_t mv;
+            if (h != v) {
+              mv = kMvLeft;
+            } else if (nin == 0 || nin == 1) {
+              mv = (bd[w] >= bu[w]) ? kMvDiag : kMvUp;
+            } else {
+              // rare multi-pred row: recover the argmax edge (first e wins)
+              const uint8_t type = (bd[w] >= bu[w]) ? kMvDiag : kMvUp;
+              const int32_t want = (type == kMvDiag) ? bd[w] : bu[w];
+              uint32_t esel = 0;
+              const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
+              for (uint32_t e = 0; e < nin; ++e) {
+                const uint32_t p = (e < kMaxPre)
+                                       ? pred_rows[e]
+                                       : c.rank[c.in_edges[node * c.ME + e]] + 1;
+                const int32_t cand =
+                    (type == kMvDiag) ? pred_val(p, j - 1) + sub : pred_val(p, j) + c.g;
+                if (cand == want) {
+                  esel = e;
+                  break;
+                }
+              }
+              mv = static_cast<uint8_t>(type | (esel << 2));
+            }
+            const int32_t h16 = h < -28000 ? -28000 : h;
+            if (store_row) {
+              Hrow[j] = static_cast<int16_t>(h16);
+            }
+            ring_row[j] = static_cast<int16_t>(h16);
+            Mrow[j] = mv;
+            if (j == len) {
+              h_sel = h;
+            }
           }
         }
-        carry_u = __builtin_amdgcn_readlane(u, kLanes - 1);
+        if (len > base && len <= base + kLanes * kWB) {
+          const int owner = static_cast<int>((len - 1 - base) / kWB);
+          const int32_t lc = __builtin_amdgcn_readlane(h_sel, owner);
+          last_col_val = lc;
+        }
       }
 
       // ring writes must be visible to every lane before the next row;
       // deliberately NOT __syncthreads (would drain the global row stores)
       wave_lds_sync();
 
-      // end-node max (strict >, first in topological order wins)
-      if (is_end) {
-        const int src_lane = static_cast<int>((len - 1) % kLanes);
-        const int32_t lc = __shfl(last_col_val, src_lane, kLanes);
-        if (lc > best_score) {
-          best_score = lc;
-          best_row = r + 1;
-        }
+      // end-node max (strict >, first in topological order wins);
+      // last_col_val is already wave-uniform (readlane)
+      if (is_end && last_col_val > best_score) {
+        best_score = last_col_val;
+        best_row = r + 1;
       }
     }
 
@@ -918,22 +1017,30 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     t_cons = lap();
     a.consensus_len[win] = clen < 0 ? 0 : static_cast<uint32_t>(clen);
     a.status[win] = c.status;
-    timing[0] = t_dp;
-    timing[1] = t_tb;
-    timing[2] = t_add;
-    timing[3] = t_topo;
-    timing[4] = t_rd;
-    timing[5] = t_cons;
-    timing[6] = wall_clock64() - t_start;
-    timing[7] = layers_done;
+    if (TIMED) {
+      timing[0] = t_dp;
+      timing[1] = t_tb;
+      timing[2] = t_add;
+      timing[3] = t_topo;
+      timing[4] = t_rd;
+      timing[5] = t_cons;
+      timing[6] = wall_clock64() - t_start;
+      timing[7] = layers_done;
+    }
   }
 }
 
 }  // namespace
 
 void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t num_windows, void* stream) {
-  hipLaunchKernelGGL(poa_window_kernel, dim3(num_windows), dim3(kLanes), 0,
-                     static_cast<hipStream_t>(stream), arena, num_windows);
+  static const bool timed = getenv("RGA_POA_TIMING") != nullptr;
+  if (timed) {
+    hipLaunchKernelGGL(poa_window_kernel<true>, dim3(num_windows), dim3(kLanes), 0,
+                       static_cast<hipStream_t>(stream), arena, num_windows);
+  } else {
+    hipLaunchKernelGGL(poa_window_kernel<false>, dim3(num_windows), dim3(kLanes), 0,
+                       static_cast<hipStream_t>(stream), arena, num_windows);
+  }
 }
 
 }  // namespace rga::hip
