@@ -104,4 +104,40 @@ def make_sample(outdir, genome_bp=20000, coverage=20, seed=0, read_len_mean=1200
         f.write("\n".join(paf_lines) + "\n")
     write_fasta(paths["layout"], [("draft0", draft.tobytes().decode())])
     write_fasta(paths["reference"], [("truth0", genome.tobytes().decode())])
+
+    # all-vs-all read overlaps (for -f fragment correction); approximate
+    # genome-projected coordinates — racon realigns the spans itself
+    ava_lines = []
+    order = np.argsort(starts)
+    min_olap = 500
+    rl = [int(r.size) for r in reads]  # actual (indel-shifted) read lengths
+    for oi in range(n_reads):
+        i = order[oi]
+        for oj in range(oi + 1, n_reads):
+            j = order[oj]
+            a = max(starts[i], starts[j])
+            b = min(starts[i] + lengths[i], starts[j] + lengths[j])
+            if b - a < min_olap:
+                if starts[j] >= starts[i] + lengths[i]:
+                    break
+                continue
+            qlen, tlen = rl[i], rl[j]
+
+            def span(s0, L, gl, rc, lo, hi):
+                # genome offsets scaled into the (indel-shifted) read
+                lo_r = min(L, (lo - s0) * L // gl)
+                hi_r = min(L, (hi - s0) * L // gl)
+                if rc:
+                    return L - hi_r, L - lo_r
+                return lo_r, hi_r
+
+            qs, qe = span(starts[i], qlen, int(lengths[i]), strands[i], a, b)
+            ts, te = span(starts[j], tlen, int(lengths[j]), strands[j], a, b)
+            rel = "-" if strands[i] != strands[j] else "+"
+            ava_lines.append(
+                f"read{i:06d}\t{qlen}\t{qs}\t{qe}\t{rel}\tread{j:06d}\t{tlen}\t{ts}\t{te}"
+                f"\t{b - a}\t{b - a}\t255")
+    paths["ava_overlaps"] = str(outdir / "ava_overlaps.paf")
+    with open(paths["ava_overlaps"], "w") as f:
+        f.write("\n".join(ava_lines) + "\n")
     return paths

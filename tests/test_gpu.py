@@ -143,3 +143,21 @@ def test_gpu_reference_sample_golden(racon, ref_data, fasta_reader):
     # Bound the HIP path to the same quality band; the exact value is pinned
     # once measured on hardware.
     assert ed < 1500, ed
+
+
+def test_gpu_fragment_correction(racon, sample):
+    """-f fragment correction through the HIP POA path (reference GPU
+    fragment goldens, racon_test.cpp:440-494): GPU output must be
+    deterministic and close to the CPU path."""
+    kw = dict(threads=4, fragment_correction=True, include_unpolished=True,
+              match=1, mismatch=-1, gap=-1)
+    cpu = racon.polish(sample["reads"], sample["ava_overlaps"], sample["reads"], **kw)
+    gpu = racon.polish(sample["reads"], sample["ava_overlaps"], sample["reads"],
+                       poa_batches=1, **kw)
+    gpu2 = racon.polish(sample["reads"], sample["ava_overlaps"], sample["reads"],
+                        poa_batches=2, **kw)
+    assert gpu == gpu2  # batch layout must not change results
+    assert len(gpu) == len(cpu)
+    total_cpu = sum(len(s) for _, s in cpu)
+    total_gpu = sum(len(s) for _, s in gpu)
+    assert abs(total_gpu - total_cpu) < 0.01 * total_cpu

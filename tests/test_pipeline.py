@@ -62,3 +62,13 @@ def test_window_length_flag_changes_output(racon, sample):
     w500 = racon.polish(sample["reads"], sample["overlaps"], sample["layout"], window_length=500)
     w1000 = racon.polish(sample["reads"], sample["overlaps"], sample["layout"], window_length=1000)
     assert w500 and w1000  # both valid; usually differ but not guaranteed
+
+
+def test_fragment_correction_cpu(racon, sample):
+    """-f all-vs-all read correction improves per-read identity (CPU)."""
+    out = racon.polish(sample["reads"], sample["ava_overlaps"], sample["reads"],
+                       threads=4, fragment_correction=True, include_unpolished=True,
+                       match=1, mismatch=-1, gap=-1)
+    assert len(out) == sample["n_reads"]
+    total = sum(len(s) for _, s in out)
+    assert total > 0.8 * 20000  # sanity: fragments kept their scale
