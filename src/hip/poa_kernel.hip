@@ -130,6 +130,7 @@ struct Shared {
   // one register bit test instead of an LDS byte read. One spare word so
   // cross-word extraction at the last chunk never reads out of bounds.
   uint64_t match[4][kMaxW / 64 + 1];
+  uint64_t rd_block[64];  // row descriptors staged 64 at a time
   uint8_t seq[kMaxW];
 };
 
@@ -631,8 +632,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     for (uint32_t j = lane; j < len; j += kLanes) {
       s.seq[j] = seq[j];
     }
-    // build the per-layer match bitvectors (lane-parallel over chunk words;
-    // one spare zero word past the end for cross-word extraction)
+    __syncthreads();
+    // build the per-layer match bitvectors from the LDS copy (lane-parallel
+    // over chunk words; one spare zero word for cross-word extraction)
     {
       const uint32_t words = (len + kLanes - 1) / kLanes + 1;
       for (uint32_t w = lane; w < words * 4; w += kLanes) {
@@ -642,7 +644,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         if (base < len) {
           const uint32_t lim = min(kLanes, len - base);
           for (uint32_t l = 0; l < lim; ++l) {
-            if (poa_code(seq[base + l]) == static_cast<int32_t>(cc)) {
+            if (poa_code(s.seq[base + l]) == static_cast<int32_t>(cc)) {
               bits |= 1ull << l;
             }
           }
@@ -665,12 +667,18 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     (void)lap();
 
     // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
-    uint64_t rd_next = c.row_desc[0];
-    for (uint32_t r = 0; r < n; ++r) {
-      const uint64_t rd = rd_next;
-      if (r + 1 < n) {
-        rd_next = c.row_desc[r + 1];  // prefetch: off the critical path
+    // Row descriptors are staged 64 at a time through LDS with one
+    // coalesced load: a per-row dependent global read (~600 ns) was the
+    // dominant per-row latency and neither scan nor store restructuring
+    // moved it (measured via RGA_POA_TIMING).
+    for (uint32_t rblk = 0; rblk < n; rblk += kLanes) {
+      if (rblk + lane < n) {
+        s.rd_block[lane] = c.row_desc[rblk + lane];
       }
+      wave_lds_sync();
+      const uint32_t rlim = min(n, rblk + kLanes);
+    for (uint32_t r = rblk; r < rlim; ++r) {
+      const uint64_t rd = s.rd_block[r - rblk];
       const uint8_t letter = static_cast<uint8_t>(rd);
       const int32_t letter_code = poa_code(letter);
       const uint32_t nin = static_cast<uint32_t>((rd >> 8) & 0xff);
@@ -924,6 +932,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         best_score = last_col_val;
         best_row = r + 1;
       }
+    }
+      wave_lds_sync();  // rows done before the next cooperative rd_block load
     }
 
     t_dp += lap();
