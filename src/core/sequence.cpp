@@ -1,14 +1,14 @@
 #include "core/sequence.hpp"
 
-#include <cctype>
-
 namespace rga {
 
 Sequence::Sequence(const char* name, uint32_t name_len, const char* data, uint32_t data_len)
     : name_(name, name_len) {
   data_.resize(data_len);
+  // branchless ASCII uppercase (auto-vectorizes; libc toupper does not)
   for (uint32_t i = 0; i < data_len; ++i) {
-    data_[i] = static_cast<char>(toupper(static_cast<unsigned char>(data[i])));
+    const char ch = data[i];
+    data_[i] = (ch >= 'a' && ch <= 'z') ? static_cast<char>(ch - 32) : ch;
   }
 }
 
