@@ -19,7 +19,7 @@ size_t slab_bytes(const PoaLimits& L) {
   b += 4 * n;                       // letters, in_cnt, out_cnt, ring_cnt
   b += n * L.max_edges * 8;         // in_edges(2) + in_weights(4) + out_edges(2)
   b += n * L.max_ring * 2;          // ring
-  b += n * 2 * 4;                   // nseq, sorted, rank, work
+  b += n * 2 * 2;                   // nseq, rank
   b += n * 12;                      // hb_score(8) + hb_pred(4)
   b += (2 * L.matrix_width + n) * 8;  // aln_nodes + aln_seq
   b += (n + 1) * L.matrix_width * 2;  // matrix
@@ -98,9 +98,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   size_t o_out_edges = carve(num_slabs_ * n * L.max_edges * 2);
   size_t o_ring = carve(num_slabs_ * n * L.max_ring * 2);
   size_t o_nseq = carve(num_slabs_ * n * 2);
-  size_t o_sorted = carve(num_slabs_ * n * 2);
   size_t o_rank = carve(num_slabs_ * n * 2);
-  size_t o_work = carve(num_slabs_ * n * 2);
   size_t o_hb_score = carve(num_slabs_ * n * 8);
   size_t o_hb_pred = carve(num_slabs_ * n * 4);
   size_t o_aln_n = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
@@ -130,9 +128,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.out_edges = reinterpret_cast<uint16_t*>(base + o_out_edges);
   arena_.ring = reinterpret_cast<uint16_t*>(base + o_ring);
   arena_.nseq = reinterpret_cast<uint16_t*>(base + o_nseq);
-  arena_.sorted = reinterpret_cast<uint16_t*>(base + o_sorted);
   arena_.rank = reinterpret_cast<uint16_t*>(base + o_rank);
-  arena_.work = reinterpret_cast<uint16_t*>(base + o_work);
   arena_.hb_score = reinterpret_cast<int64_t*>(base + o_hb_score);
   arena_.hb_pred = reinterpret_cast<int32_t*>(base + o_hb_pred);
   arena_.aln_nodes = reinterpret_cast<int32_t*>(base + o_aln_n);

@@ -157,9 +157,7 @@ struct WindowCtx {
   uint16_t* out_edges;
   uint16_t* ring;
   uint16_t* nseq;
-  uint16_t* sorted;
   uint16_t* rank;
-  uint16_t* work;
   int64_t* hb_score;
   int32_t* hb_pred;
   int32_t* aln_nodes;
@@ -543,9 +541,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
   c.out_edges = a.out_edges + static_cast<size_t>(slab) * L.max_nodes * L.max_edges;
   c.ring = a.ring + static_cast<size_t>(slab) * L.max_nodes * L.max_ring;
   c.nseq = a.nseq + static_cast<size_t>(slab) * L.max_nodes;
-  c.sorted = a.sorted + static_cast<size_t>(slab) * L.max_nodes;
   c.rank = a.rank + static_cast<size_t>(slab) * L.max_nodes;
-  c.work = a.work + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_score = a.hb_score + static_cast<size_t>(slab) * L.max_nodes;
   c.hb_pred = a.hb_pred + static_cast<size_t>(slab) * L.max_nodes;
   c.aln_nodes = a.aln_nodes + static_cast<size_t>(slab) * (2 * L.matrix_width + L.max_nodes);

@@ -61,9 +61,8 @@ struct PoaDeviceArena {
   uint16_t* out_edges;   // [max_nodes * max_edges]
   uint16_t* ring;        // [max_nodes * max_ring]
   uint16_t* nseq;        // [max_nodes] sequences touching node (coverage)
-  uint16_t* sorted;      // [max_nodes] topological order
-  uint16_t* rank;        // [max_nodes] node -> rank
-  uint16_t* work;        // [max_nodes] Kahn queue / in-degree scratch
+  uint16_t* rank;        // [max_nodes] node -> rank (Kahn scratch/queue and
+                         // the topological order itself live in LDS)
   int64_t* hb_score;     // [max_nodes] heaviest-bundle running scores
   int32_t* hb_pred;      // [max_nodes]
   int32_t* aln_nodes;    // [2 * matrix_width + max_nodes] alignment node ids
