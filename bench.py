@@ -51,6 +51,8 @@ def main():
                     help="CPU threads per rank (default: ncpu / (2*world))")
     ap.add_argument("--poa-batches", type=int, default=8)
     ap.add_argument("--aligner-batches", type=int, default=4)
+    ap.add_argument("--banded", action="store_true",
+                    help="use -b static-band POA (approximation; off by default)")
     ap.add_argument("--cpu", action="store_true", help="force CPU path (debug)")
     args = ap.parse_args()
 
@@ -97,7 +99,8 @@ def main():
     def one_step():
         out = _racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
                             threads=args.threads, window_length=args.window,
-                            poa_batches=poa_batches, aligner_batches=aligner_batches)
+                            poa_batches=poa_batches, aligner_batches=aligner_batches,
+                            banded_poa=args.banded)
         # gather polished contigs to rank 0 (variable-length bytes over RCCL)
         if world > 1:
             blob = "".join(s for _, s in out).encode()
