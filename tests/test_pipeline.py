@@ -72,3 +72,17 @@ def test_fragment_correction_cpu(racon, sample):
     assert len(out) == sample["n_reads"]
     total = sum(len(s) for _, s in out)
     assert total > 0.8 * 20000  # sanity: fragments kept their scale
+
+
+def test_ngs_mode_short_reads(racon, tmp_path):
+    """Mean read length <= 1000 flips WindowType to kNGS (no consensus trim,
+    reference polisher.cpp:276-277). The pipeline must run and still correct
+    the draft."""
+    from racon_amd import synth
+    s = synth.make_sample(tmp_path, genome_bp=8000, coverage=25, seed=11,
+                          read_len_mean=600, read_len_sd=50)
+    out = racon.polish(s["reads"], s["overlaps"], s["layout"], threads=2)
+    assert len(out) == 1
+    truth = open(s["reference"]).read().splitlines()[1]
+    draft = open(s["layout"]).read().splitlines()[1]
+    assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth)
