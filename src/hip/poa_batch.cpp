@@ -3,6 +3,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <chrono>
 #include <cstring>
 
 #include "hip/hip_common.hpp"
@@ -255,6 +256,9 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   if (windows_.empty()) {
     return polished;
   }
+  using clk = std::chrono::steady_clock;
+  const bool host_timing = getenv("RGA_POA_HOSTTIME") != nullptr;
+  auto t0 = clk::now();
 
   pack();
   const size_t nw0 = windows_.size();
@@ -301,7 +305,9 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   RGA_HIP_CHECK(hipMemcpyAsync(h_consensus_len_, arena_.consensus_len, nw * 4,
                                hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, nw * 4, hipMemcpyDeviceToHost, s));
+  auto t1 = clk::now();
   RGA_HIP_CHECK(hipStreamSynchronize(s));
+  auto t2 = clk::now();
 
   if (getenv("RGA_POA_TIMING") != nullptr) {
     std::vector<unsigned long long> t(nw * 8);
@@ -354,6 +360,16 @@ std::vector<bool> PoaBatch::generate(bool trim) {
       window->set_consensus(std::move(consensus));
     }
     polished[orig] = status;
+  }
+  if (host_timing) {
+    auto t3 = clk::now();
+    auto ms = [](auto a, auto b) {
+      return std::chrono::duration_cast<std::chrono::microseconds>(b - a).count() / 1000.0;
+    };
+    fprintf(stderr,
+            "[rga::hip::PoaBatch] host timing (%zu windows): pack+sort+h2d+launch %.1f ms, "
+            "gpu sync %.1f ms, post %.1f ms\n",
+            nw, ms(t0, t1), ms(t1, t2), ms(t2, t3));
   }
   return polished;
 }
