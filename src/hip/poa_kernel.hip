@@ -810,7 +810,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           for (uint32_t e = 0; e < nin; ++e) {
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
                                              : c.rank[c.in_edges[node * c.ME + e]] + 1;
-            // gather pred row values pv[w] = H(p, cbase + w), w in 0..kWB
+            // gather pred row values pv[w] = H(p, cbase + w), w in 0..kWB.
+            // Loads are UNCONDITIONAL with a clamped index + VALU select:
+            // per-element predication compiled to one exec-branched
+            // flat_load each (generic pointer), serializing the row. The
+            // LDS and global paths are separate loops so each keeps its
+            // address space (ds_read vs global_load).
             int32_t pv[kWB + 1];
             if (p == 0) {
 #pragma unroll
@@ -827,15 +832,27 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
                 phi = min(len, (pkhi + 1) * kLanes);
                 p_has0 = (pklo == 0);
               }
-              const int16_t* src = (r + 1 - p < kRing)
-                                       ? s.u.ring[p % kRing]
-                                       : c.matrix + static_cast<size_t>(p) * c.MW;
+              const uint32_t colmax = c.MW - 1;
+              if (r + 1 - p < kRing) {
+                const uint32_t slot = p % kRing;
 #pragma unroll
-              for (uint32_t w = 0; w <= kWB; ++w) {
-                const uint32_t col = cbase + w;
-                const bool ok =
-                    (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
-                pv[w] = ok ? static_cast<int32_t>(src[col]) : kNegInf;
+                for (uint32_t w = 0; w <= kWB; ++w) {
+                  const uint32_t col = cbase + w;
+                  const int32_t val = s.u.ring[slot][min(col, colmax)];
+                  const bool ok =
+                      (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
+                  pv[w] = ok ? val : kNegInf;
+                }
+              } else {
+                const int16_t* gsrc = c.matrix + static_cast<size_t>(p) * c.MW;
+#pragma unroll
+                for (uint32_t w = 0; w <= kWB; ++w) {
+                  const uint32_t col = cbase + w;
+                  const int32_t val = gsrc[min(col, colmax)];
+                  const bool ok =
+                      (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
+                  pv[w] = ok ? val : kNegInf;
+                }
               }
             }
 #pragma unroll
