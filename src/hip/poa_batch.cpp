@@ -313,13 +313,18 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     std::vector<unsigned long long> t(nw * 8);
     RGA_HIP_CHECK(hipMemcpy(t.data(), arena_.timing, nw * 8 * 8, hipMemcpyDeviceToHost));
     unsigned long long sum[8] = {0};
+    std::vector<unsigned long long> totals(nw);
     for (size_t i = 0; i < nw; ++i) {
       for (int k = 0; k < 8; ++k) sum[k] += t[i * 8 + k];
+      totals[i] = t[i * 8 + 6];
     }
+    std::sort(totals.begin(), totals.end());
     fprintf(stderr,
             "[rga::hip::PoaBatch] timing (wall ticks, %zu windows): dp=%llu tb=%llu "
-            "add=%llu topo=%llu rdesc=%llu cons=%llu total=%llu layers=%llu\n",
-            nw, sum[0], sum[1], sum[2], sum[3], sum[4], sum[5], sum[6], sum[7]);
+            "add=%llu topo=%llu rdesc=%llu cons=%llu total=%llu layers=%llu\n"
+            "[rga::hip::PoaBatch] window total ticks: p50=%llu p90=%llu p99=%llu max=%llu\n",
+            nw, sum[0], sum[1], sum[2], sum[3], sum[4], sum[5], sum[6], sum[7],
+            totals[nw / 2], totals[nw * 9 / 10], totals[nw * 99 / 100], totals[nw - 1]);
   }
 
   // CPU-parity post-processing (reference cudabatch.cpp:199-261).
