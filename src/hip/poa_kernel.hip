@@ -27,7 +27,6 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdlib>
-#include <type_traits>
 
 #include "hip/poa_types.hpp"
 
@@ -668,13 +667,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
     // coalesced load: a per-row dependent global read (~600 ns) was the
     // dominant per-row latency and neither scan nor store restructuring
     // moved it (measured via RGA_POA_TIMING).
-    //
-    // The whole row loop is templated on WB = columns per lane, picked per
-    // layer as the smallest even count covering the row window in ONE pass:
-    // a fixed kWB=8 made a 530-column window cost two full wave passes (the
-    // second nearly empty) — ~1.9x wasted issue for typical 500 bp windows.
-    auto run_rows = [&](auto wbc) {
-      constexpr uint32_t WB = decltype(wbc)::value;
     for (uint32_t rblk = 0; rblk < n; rblk += kLanes) {
       if (rblk + lane < n) {
         s.rd_block[lane] = c.row_desc[rblk + lane];
@@ -766,7 +758,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       }
 
       // ---- lane-blocked columns ----
-      // Lane l owns WB contiguous columns per pass; one register-local
+      // Lane l owns kWB contiguous columns per pass; one register-local
       // inclusive scan + one DPP wave scan per pass replaces the previous
       // chunk-serial carry chain (8 dependent LDS+scan segments per row).
       const uint32_t j0 = banded ? row_klo * kLanes : 0;
@@ -775,10 +767,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
                              // the band excludes column 0)
       int32_t last_col_val = kNegInf;
 
-      for (uint32_t base = j0; base < jend; base += kLanes * WB) {
-        const uint32_t cbase = base + lane * WB;  // own cols: cbase+1..cbase+WB
+      for (uint32_t base = j0; base < jend; base += kLanes * kWB) {
+        const uint32_t cbase = base + lane * kWB;  // own cols: cbase+1..cbase+kWB
         const uint32_t nown =
-            (cbase < jend) ? min(WB, jend - cbase) : 0;
+            (cbase < jend) ? min(kWB, jend - cbase) : 0;
 
         // substitution matches for own columns: bits w of mbits
         uint64_t mbits = 0;
@@ -797,16 +789,16 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           }
         }
 
-        int32_t bd[WB], bu[WB];  // best diagonal / vertical candidates
+        int32_t bd[kWB], bu[kWB];  // best diagonal / vertical candidates
 #pragma unroll
-        for (uint32_t w = 0; w < WB; ++w) {
+        for (uint32_t w = 0; w < kWB; ++w) {
           bd[w] = kNegInf;
           bu[w] = kNegInf;
         }
 
         if (nin == 0) {
 #pragma unroll
-          for (uint32_t w = 0; w < WB; ++w) {
+          for (uint32_t w = 0; w < kWB; ++w) {
             if (w < nown) {
               const int32_t j = static_cast<int32_t>(cbase + 1 + w);
               const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
@@ -818,16 +810,16 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           for (uint32_t e = 0; e < nin; ++e) {
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
                                              : c.rank[c.in_edges[node * c.ME + e]] + 1;
-            // gather pred row values pv[w] = H(p, cbase + w), w in 0..WB.
+            // gather pred row values pv[w] = H(p, cbase + w), w in 0..kWB.
             // Loads are UNCONDITIONAL with a clamped index + VALU select:
             // per-element predication compiled to one exec-branched
             // flat_load each (generic pointer), serializing the row. The
             // LDS and global paths are separate loops so each keeps its
             // address space (ds_read vs global_load).
-            int32_t pv[WB + 1];
+            int32_t pv[kWB + 1];
             if (p == 0) {
 #pragma unroll
-              for (uint32_t w = 0; w <= WB; ++w) {
+              for (uint32_t w = 0; w <= kWB; ++w) {
                 pv[w] = static_cast<int32_t>(cbase + w) * c.g;
               }
             } else {
@@ -844,7 +836,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               if (r + 1 - p < kRing) {
                 const uint32_t slot = p % kRing;
 #pragma unroll
-                for (uint32_t w = 0; w <= WB; ++w) {
+                for (uint32_t w = 0; w <= kWB; ++w) {
                   const uint32_t col = cbase + w;
                   const int32_t val = s.u.ring[slot][min(col, colmax)];
                   const bool ok =
@@ -854,7 +846,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               } else {
                 const int16_t* gsrc = c.matrix + static_cast<size_t>(p) * c.MW;
 #pragma unroll
-                for (uint32_t w = 0; w <= WB; ++w) {
+                for (uint32_t w = 0; w <= kWB; ++w) {
                   const uint32_t col = cbase + w;
                   const int32_t val = gsrc[min(col, colmax)];
                   const bool ok =
@@ -864,7 +856,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               }
             }
 #pragma unroll
-            for (uint32_t w = 0; w < WB; ++w) {
+            for (uint32_t w = 0; w < kWB; ++w) {
               if (w < nown) {
                 const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
                 bd[w] = max(bd[w], pv[w] + sub);
@@ -875,10 +867,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         }
 
         // local inclusive u-space scan over own columns
-        int32_t us[WB];
+        int32_t us[kWB];
         int32_t run = kNegInf;
 #pragma unroll
-        for (uint32_t w = 0; w < WB; ++w) {
+        for (uint32_t w = 0; w < kWB; ++w) {
           const int32_t j = static_cast<int32_t>(cbase + 1 + w);
           const int32_t v = max(bd[w], bu[w]);
           const int32_t u = (w < nown) ? v - j * c.g : kNegInf;
@@ -896,7 +888,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         // finalize own columns: h, moves, stores
         int32_t h_sel = kNegInf;
 #pragma unroll
-        for (uint32_t w = 0; w < WB; ++w) {
+        for (uint32_t w = 0; w < kWB; ++w) {
           if (w < nown) {
             const uint32_t j = cbase + 1 + w;
             const int32_t v = max(bd[w], bu[w]);
@@ -936,8 +928,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
             }
           }
         }
-        if (len > base && len <= base + kLanes * WB) {
-          const int owner = static_cast<int>((len - 1 - base) / WB);
+        if (len > base && len <= base + kLanes * kWB) {
+          const int owner = static_cast<int>((len - 1 - base) / kWB);
           const int32_t lc = __builtin_amdgcn_readlane(h_sel, owner);
           last_col_val = lc;
         }
@@ -955,27 +947,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       }
     }
       wave_lds_sync();  // rows done before the next cooperative rd_block load
-    }
-    };  // run_rows
-
-    {
-      const uint32_t aw_max = banded ? min(len, c.bw + kLanes) : len;
-      const uint32_t wneed = (aw_max + kLanes - 1) / kLanes;
-      if (wneed <= 2) {
-        run_rows(std::integral_constant<uint32_t, 2>{});
-      } else if (wneed <= 4) {
-        run_rows(std::integral_constant<uint32_t, 4>{});
-      } else if (wneed <= 6) {
-        run_rows(std::integral_constant<uint32_t, 6>{});
-      } else if (wneed <= 8) {
-        run_rows(std::integral_constant<uint32_t, 8>{});
-      } else if (wneed <= 10) {
-        run_rows(std::integral_constant<uint32_t, 10>{});
-      } else if (wneed <= 12) {
-        run_rows(std::integral_constant<uint32_t, 12>{});
-      } else {
-        run_rows(std::integral_constant<uint32_t, 16>{});
-      }
     }
 
     t_dp += lap();
