@@ -161,3 +161,38 @@ def test_gpu_fragment_correction(racon, sample):
     total_cpu = sum(len(s) for _, s in cpu)
     total_gpu = sum(len(s) for _, s in gpu)
     assert abs(total_gpu - total_cpu) < 0.01 * total_cpu
+
+
+def test_gpu_pinned_goldens(racon, sample):
+    """Exact pinned GPU outputs on the fixed synthetic sample (the reference
+    pins exact CUDA goldens the same way, racon_test.cpp:292-496). Regenerate
+    with RGA_PIN=1 after an intentional kernel-semantics change."""
+    import hashlib
+    import json
+    import os
+    from pathlib import Path
+
+    golden_path = Path(__file__).parent / "goldens_gpu.json"
+
+    polished = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
+                            threads=4, poa_batches=1, aligner_batches=1)
+    h_polish = hashlib.sha256(("".join(s for _, s in polished)).encode()).hexdigest()
+
+    frags = racon.polish(sample["reads"], sample["ava_overlaps"], sample["reads"],
+                         threads=4, poa_batches=1, fragment_correction=True,
+                         include_unpolished=True, match=1, mismatch=-1, gap=-1)
+    h_frag = hashlib.sha256(("".join(s for _, s in frags)).encode()).hexdigest()
+
+    got = {"polish_sha256": h_polish, "fragment_sha256": h_frag,
+           "polish_len": sum(len(s) for _, s in polished),
+           "fragment_len": sum(len(s) for _, s in frags)}
+
+    if os.environ.get("RGA_PIN"):
+        golden_path.write_text(json.dumps(got, indent=1) + "\n")
+        print(f"pinned: {got}")
+        return
+    if not golden_path.exists():
+        import pytest
+        pytest.skip("no pinned GPU goldens yet (run with RGA_PIN=1 on a GPU host)")
+    want = json.loads(golden_path.read_text())
+    assert got == want
