@@ -186,6 +186,7 @@ bool PoaBatch::add_window(const std::shared_ptr<Window>& window, bool* never_fit
 
   // count + measure what ships (skip too-long layers, cap depth)
   uint32_t shipped = 1;
+  uint32_t max_len = window->sequence(0).second;
   size_t bytes = window->sequence(0).second;
   for (uint32_t k = 1; k < total_layers && shipped < max_depth_ + 1; ++k) {
     uint32_t i = order[k];
@@ -194,6 +195,7 @@ bool PoaBatch::add_window(const std::shared_ptr<Window>& window, bool* never_fit
       continue;  // reference: exceeded_maximum_sequence_size -> skipped
     }
     bytes += len;
+    max_len = std::max(max_len, len);
     ++shipped;
   }
 
@@ -207,6 +209,7 @@ bool PoaBatch::add_window(const std::shared_ptr<Window>& window, bool* never_fit
   desc.seq_offset = static_cast<uint32_t>(seq_bytes_);
   desc.scratch_idx = win_idx;
   desc.num_seqs = shipped;
+  desc.max_len = max_len;
   h_layer_index_[win_idx] = static_cast<uint32_t>(num_layer_ends_);
   seq_bytes_ += bytes;
   num_layer_ends_ += shipped;
@@ -262,15 +265,27 @@ std::vector<bool> PoaBatch::generate(bool trim) {
 
   pack();
   const size_t nw0 = windows_.size();
-  // dispatch heaviest windows first: blocks launch roughly in order, so the
-  // long poles start early instead of defining the tail of the launch
+  // order: columns-per-lane bucket first (each bucket is its own kernel
+  // instantiation launched over a contiguous desc range), heaviest windows
+  // first within a bucket (blocks launch roughly in order, so the long
+  // poles start early instead of defining the tail)
   std::vector<uint32_t> perm(nw0);
   for (uint32_t i = 0; i < nw0; ++i) perm[i] = i;
+  auto bucket = [&](uint32_t w) -> uint32_t {
+    uint32_t aw = h_desc_[w].max_len;
+    if (arena_.band_width != 0) {
+      aw = std::min(aw, arena_.band_width + 64);
+    }
+    const uint32_t need = (aw + 63) / 64;
+    return need <= 5 ? 5u : (need <= 9 ? 9u : 16u);
+  };
   auto cost = [&](uint32_t w) {
     const uint32_t first = h_layer_index_[w];
     return h_layer_ends_[first + h_desc_[w].num_seqs - 1];  // total layer bytes
   };
   std::sort(perm.begin(), perm.end(), [&](uint32_t a, uint32_t b) {
+    const uint32_t ba = bucket(a), bb = bucket(b);
+    if (ba != bb) return ba < bb;
     const uint32_t ca = cost(a), cb = cost(b);
     if (ca != cb) return ca > cb;
     return a < b;
@@ -295,7 +310,19 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   d(arena_.layer_ends_index, h_layer_index_, windows_.size() * 4);
   d(arena_.windows, h_desc_, windows_.size() * sizeof(PoaWindowDesc));
 
-  launch_poa_kernel(arena_, static_cast<uint32_t>(windows_.size()), stream_);
+  // one launch per columns-per-lane bucket over its contiguous range
+  {
+    uint32_t begin = 0;
+    while (begin < nw0) {
+      const uint32_t wb = bucket(perm[begin]);
+      uint32_t end = begin + 1;
+      while (end < nw0 && bucket(perm[end]) == wb) {
+        ++end;
+      }
+      launch_poa_kernel(arena_, begin, end - begin, wb, stream_);
+      begin = end;
+    }
+  }
 
   const size_t nw = windows_.size();
   RGA_HIP_CHECK(hipMemcpyAsync(h_consensus_, arena_.consensus,

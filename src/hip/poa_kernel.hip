@@ -40,7 +40,6 @@ constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
 constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
 constexpr uint32_t kMaxPre = 4;   // predecessor rows precomputed per row
-constexpr uint32_t kWB = 8;       // contiguous columns per lane per DP pass
 
 // move byte encoding
 constexpr uint8_t kMvDiag = 0;
@@ -516,14 +515,14 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
 
 // ---------- the mega-kernel ----------
 
-template <bool TIMED>
-__attribute__((amdgpu_waves_per_eu(5)))
+template <bool TIMED, uint32_t WB>
 __launch_bounds__(kLanes)
-__global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
-  const uint32_t win = blockIdx.x;
-  if (win >= num_windows) {
+__global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
+                                  uint32_t num_windows) {
+  if (blockIdx.x >= num_windows) {
     return;
   }
+  const uint32_t win = window_base + blockIdx.x;
   const int lane = threadIdx.x;
   const PoaWindowDesc desc = a.windows[win];
   const PoaLimits L = a.limits;
@@ -758,7 +757,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
       }
 
       // ---- lane-blocked columns ----
-      // Lane l owns kWB contiguous columns per pass; one register-local
+      // Lane l owns WB contiguous columns per pass; one register-local
       // inclusive scan + one DPP wave scan per pass replaces the previous
       // chunk-serial carry chain (8 dependent LDS+scan segments per row).
       const uint32_t j0 = banded ? row_klo * kLanes : 0;
@@ -767,10 +766,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
                              // the band excludes column 0)
       int32_t last_col_val = kNegInf;
 
-      for (uint32_t base = j0; base < jend; base += kLanes * kWB) {
-        const uint32_t cbase = base + lane * kWB;  // own cols: cbase+1..cbase+kWB
+      for (uint32_t base = j0; base < jend; base += kLanes * WB) {
+        const uint32_t cbase = base + lane * WB;  // own cols: cbase+1..cbase+WB
         const uint32_t nown =
-            (cbase < jend) ? min(kWB, jend - cbase) : 0;
+            (cbase < jend) ? min(WB, jend - cbase) : 0;
 
         // substitution matches for own columns: bits w of mbits
         uint64_t mbits = 0;
@@ -789,16 +788,16 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           }
         }
 
-        int32_t bd[kWB], bu[kWB];  // best diagonal / vertical candidates
+        int32_t bd[WB], bu[WB];  // best diagonal / vertical candidates
 #pragma unroll
-        for (uint32_t w = 0; w < kWB; ++w) {
+        for (uint32_t w = 0; w < WB; ++w) {
           bd[w] = kNegInf;
           bu[w] = kNegInf;
         }
 
         if (nin == 0) {
 #pragma unroll
-          for (uint32_t w = 0; w < kWB; ++w) {
+          for (uint32_t w = 0; w < WB; ++w) {
             if (w < nown) {
               const int32_t j = static_cast<int32_t>(cbase + 1 + w);
               const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
@@ -810,16 +809,16 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
           for (uint32_t e = 0; e < nin; ++e) {
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
                                              : c.rank[c.in_edges[node * c.ME + e]] + 1;
-            // gather pred row values pv[w] = H(p, cbase + w), w in 0..kWB.
+            // gather pred row values pv[w] = H(p, cbase + w), w in 0..WB.
             // Loads are UNCONDITIONAL with a clamped index + VALU select:
             // per-element predication compiled to one exec-branched
             // flat_load each (generic pointer), serializing the row. The
             // LDS and global paths are separate loops so each keeps its
             // address space (ds_read vs global_load).
-            int32_t pv[kWB + 1];
+            int32_t pv[WB + 1];
             if (p == 0) {
 #pragma unroll
-              for (uint32_t w = 0; w <= kWB; ++w) {
+              for (uint32_t w = 0; w <= WB; ++w) {
                 pv[w] = static_cast<int32_t>(cbase + w) * c.g;
               }
             } else {
@@ -836,7 +835,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               if (r + 1 - p < kRing) {
                 const uint32_t slot = p % kRing;
 #pragma unroll
-                for (uint32_t w = 0; w <= kWB; ++w) {
+                for (uint32_t w = 0; w <= WB; ++w) {
                   const uint32_t col = cbase + w;
                   const int32_t val = s.u.ring[slot][min(col, colmax)];
                   const bool ok =
@@ -846,7 +845,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               } else {
                 const int16_t* gsrc = c.matrix + static_cast<size_t>(p) * c.MW;
 #pragma unroll
-                for (uint32_t w = 0; w <= kWB; ++w) {
+                for (uint32_t w = 0; w <= WB; ++w) {
                   const uint32_t col = cbase + w;
                   const int32_t val = gsrc[min(col, colmax)];
                   const bool ok =
@@ -856,7 +855,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
               }
             }
 #pragma unroll
-            for (uint32_t w = 0; w < kWB; ++w) {
+            for (uint32_t w = 0; w < WB; ++w) {
               if (w < nown) {
                 const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
                 bd[w] = max(bd[w], pv[w] + sub);
@@ -867,10 +866,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         }
 
         // local inclusive u-space scan over own columns
-        int32_t us[kWB];
+        int32_t us[WB];
         int32_t run = kNegInf;
 #pragma unroll
-        for (uint32_t w = 0; w < kWB; ++w) {
+        for (uint32_t w = 0; w < WB; ++w) {
           const int32_t j = static_cast<int32_t>(cbase + 1 + w);
           const int32_t v = max(bd[w], bu[w]);
           const int32_t u = (w < nown) ? v - j * c.g : kNegInf;
@@ -888,7 +887,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
         // finalize own columns: h, moves, stores
         int32_t h_sel = kNegInf;
 #pragma unroll
-        for (uint32_t w = 0; w < kWB; ++w) {
+        for (uint32_t w = 0; w < WB; ++w) {
           if (w < nown) {
             const uint32_t j = cbase + 1 + w;
             const int32_t v = max(bd[w], bu[w]);
@@ -928,8 +927,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
             }
           }
         }
-        if (len > base && len <= base + kLanes * kWB) {
-          const int owner = static_cast<int>((len - 1 - base) / kWB);
+        if (len > base && len <= base + kLanes * WB) {
+          const int owner = static_cast<int>((len - 1 - base) / WB);
           const int32_t lc = __builtin_amdgcn_readlane(h_sel, owner);
           last_col_val = lc;
         }
@@ -1055,14 +1054,44 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t num_windows) {
 
 }  // namespace
 
-void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t num_windows, void* stream) {
+void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
+                       uint32_t num_windows, uint32_t wb, void* stream) {
   static const bool timed = getenv("RGA_POA_TIMING") != nullptr;
+  auto st = static_cast<hipStream_t>(stream);
+  const dim3 grid(num_windows), block(kLanes);
+  // separate __global__ instantiations per columns-per-lane bucket: one
+  // kernel containing all variants pays the widest variant's registers on
+  // every path (measured 3x occupancy collapse), separate kernels do not
   if (timed) {
-    hipLaunchKernelGGL(poa_window_kernel<true>, dim3(num_windows), dim3(kLanes), 0,
-                       static_cast<hipStream_t>(stream), arena, num_windows);
+    switch (wb) {
+      case 5:
+        hipLaunchKernelGGL((poa_window_kernel<true, 5>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 9:
+        hipLaunchKernelGGL((poa_window_kernel<true, 9>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      default:
+        hipLaunchKernelGGL((poa_window_kernel<true, 16>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+    }
   } else {
-    hipLaunchKernelGGL(poa_window_kernel<false>, dim3(num_windows), dim3(kLanes), 0,
-                       static_cast<hipStream_t>(stream), arena, num_windows);
+    switch (wb) {
+      case 5:
+        hipLaunchKernelGGL((poa_window_kernel<false, 5>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 9:
+        hipLaunchKernelGGL((poa_window_kernel<false, 9>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      default:
+        hipLaunchKernelGGL((poa_window_kernel<false, 16>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+    }
   }
 }
 

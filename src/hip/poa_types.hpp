@@ -40,6 +40,8 @@ struct PoaWindowDesc {
   uint32_t seq_offset;   // byte offset into the packed seq/weight arena
   uint32_t num_seqs;     // layers shipped to the device (<= max_depth + 1)
   uint32_t scratch_idx;  // which device slab this window uses
+  uint32_t max_len;      // longest layer (columns) — selects the kernel's
+                         // columns-per-lane instantiation
 };
 
 // Device arena pointers (one allocation, carved into slabs).
@@ -93,6 +95,10 @@ struct PoaDeviceArena {
   PoaLimits limits;
 };
 
-void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t num_windows, void* stream);
+// Launches the columns-per-lane kernel variant for windows
+// [window_base, window_base + num_windows) of the (bucket-sorted) desc
+// array. wb must be 5, 9 or 16 and cover ceil(window max_len / 64).
+void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
+                       uint32_t num_windows, uint32_t wb, void* stream);
 
 }  // namespace rga::hip
