@@ -277,7 +277,7 @@ std::vector<bool> PoaBatch::generate(bool trim) {
       aw = std::min(aw, arena_.band_width + 64);
     }
     const uint32_t need = (aw + 63) / 64;
-    return need <= 5 ? 5u : (need <= 9 ? 9u : 16u);
+    return need <= 5 ? 5u : 8u;  // 8-wide multi-pass covers any width
   };
   auto cost = [&](uint32_t w) {
     const uint32_t first = h_layer_index_[w];

@@ -1067,18 +1067,18 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
   // separate __global__ instantiations per columns-per-lane bucket: one
   // kernel containing all variants pays the widest variant's registers on
   // every path (measured 3x occupancy collapse), separate kernels do not
+  // Measured: the power-of-two 8-wide variant (two passes for a 530-column
+  // window) beats a 9-wide single-pass variant by ~8% — non-power-of-two
+  // unrolls lose more in generated address code than the extra, mostly
+  // empty pass costs. 5-wide serves the banded (-b) window.
   if (timed) {
     switch (wb) {
       case 5:
         hipLaunchKernelGGL((poa_window_kernel<true, 5>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
-      case 9:
-        hipLaunchKernelGGL((poa_window_kernel<true, 9>), grid, block, 0, st, arena,
-                           window_base, num_windows);
-        break;
       default:
-        hipLaunchKernelGGL((poa_window_kernel<true, 16>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<true, 8>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
     }
@@ -1088,12 +1088,8 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
         hipLaunchKernelGGL((poa_window_kernel<false, 5>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
-      case 9:
-        hipLaunchKernelGGL((poa_window_kernel<false, 9>), grid, block, 0, st, arena,
-                           window_base, num_windows);
-        break;
       default:
-        hipLaunchKernelGGL((poa_window_kernel<false, 16>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<false, 8>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
     }
