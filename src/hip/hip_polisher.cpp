@@ -137,11 +137,14 @@ class HipPolisher : public Polisher {
         }
       }
       if (count > 0) {
-        // 10% of mean span like the reference, clamped: the Myers band is
-        // exact inside the band and the CPU fallback catches escapes, so a
-        // tight band wins (see pick_band_k in aligner_batch.cpp)
+        // 10% of mean span like the reference, clamped to 256 (K=4): the
+        // Myers band is exact inside the band, indel drift on long reads is
+        // a ~sqrt random walk (sd ~30 bp at 20 kbp / 6% error, so ±128 is
+        // >4 sigma), the CPU fallback catches escapes per item — and the
+        // aligner launch is only ~1.5 waves/CU, so per-wave work is wall
+        // time one-for-one
         band = static_cast<uint32_t>(total_len / count / 10) & ~1u;
-        band = band < 256 ? 256 : (band > 512 ? 512 : band);
+        band = band < 64 ? 64 : (band > 256 ? 256 : band);
       }
     }
 
