@@ -3,9 +3,12 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <atomic>
 #include <cstring>
 #include <numeric>
 #include <string>
+#include <thread>
+#include <vector>
 
 #include "hip/hip_common.hpp"
 
@@ -295,19 +298,43 @@ uint32_t AlignerBatch::align_and_emit() {
     return 0;
   }
   run();
-  uint32_t failed = 0;
-  for (size_t i = 0; i < overlaps_.size(); ++i) {
-    if (overlaps_[i] == nullptr) {
-      continue;
+  // CIGAR strings are independent per slot: build them on a few threads
+  // (the run-length walk over ~30 kbp paths x thousands of alignments is
+  // otherwise a serial tail after every batch round)
+  const size_t n = overlaps_.size();
+  const uint32_t nthreads = std::min<uint32_t>(4, std::max<uint32_t>(1, n / 512));
+  std::atomic<uint32_t> failed{0};
+  auto emit_range = [&](size_t begin, size_t end) {
+    uint32_t local_failed = 0;
+    for (size_t i = begin; i < end; ++i) {
+      if (overlaps_[i] == nullptr) {
+        continue;
+      }
+      std::string cigar = cigar_of(static_cast<uint32_t>(i));
+      if (cigar.empty()) {
+        ++local_failed;  // empty CIGAR -> CPU pairwise fallback
+        continue;
+      }
+      overlaps_[i]->set_cigar(cigar);
     }
-    std::string cigar = cigar_of(static_cast<uint32_t>(i));
-    if (cigar.empty()) {
-      ++failed;  // empty CIGAR -> CPU pairwise fallback
-      continue;
+    failed += local_failed;
+  };
+  if (nthreads <= 1) {
+    emit_range(0, n);
+  } else {
+    std::vector<std::thread> threads;
+    const size_t step = (n + nthreads - 1) / nthreads;
+    for (uint32_t t = 0; t < nthreads; ++t) {
+      const size_t b = t * step, e = std::min(n, b + step);
+      if (b < e) {
+        threads.emplace_back(emit_range, b, e);
+      }
     }
-    overlaps_[i]->set_cigar(cigar);
+    for (auto& t : threads) {
+      t.join();
+    }
   }
-  return failed;
+  return failed.load();
 }
 
 void AlignerBatch::reset() {
