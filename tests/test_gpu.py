@@ -196,3 +196,29 @@ def test_gpu_pinned_goldens(racon, sample):
         pytest.skip("no pinned GPU goldens yet (run with RGA_PIN=1 on a GPU host)")
     want = json.loads(golden_path.read_text())
     assert got == want
+
+
+def test_gpu_stress_configs(racon, tmp_path_factory, fasta_reader):
+    """Edge configs: depth near the 200-layer cap, and 12%-error reads that
+    stress the aligner band (escapes must fall back per item, and output
+    quality must still improve the draft)."""
+    from racon_amd import synth
+
+    # deep coverage: windows hit MAX_DEPTH_PER_WINDOW
+    d1 = tmp_path_factory.mktemp("deep")
+    s1 = synth.make_sample(d1, genome_bp=15000, coverage=250, seed=21)
+    out = racon.polish(s1["reads"], s1["overlaps"], s1["layout"],
+                       threads=4, poa_batches=1, aligner_batches=1)
+    truth = list(fasta_reader(s1["reference"]).values())[0]
+    draft = list(fasta_reader(s1["layout"]).values())[0]
+    assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth) * 0.2
+
+    # high error: ~12% total error rate
+    d2 = tmp_path_factory.mktemp("noisy")
+    s2 = synth.make_sample(d2, genome_bp=30000, coverage=40, seed=22,
+                           sub=0.04, ins=0.04, dele=0.04)
+    out = racon.polish(s2["reads"], s2["overlaps"], s2["layout"],
+                       threads=4, poa_batches=1, aligner_batches=1)
+    truth = list(fasta_reader(s2["reference"]).values())[0]
+    draft = list(fasta_reader(s2["layout"]).values())[0]
+    assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth)
