@@ -99,3 +99,17 @@ def test_cudapoa_batches_optional_arg(racon_cli, sample):
                                 sample["layout"]])
     assert out.returncode != 0
     assert "missing input file" in out.stderr
+
+
+@pytest.mark.parametrize("name,reads,ovl,tgt,msg", [
+    ("bad_paf", ">a\nACGT\n", "read1\tnotanum\t0\n", ">a\nACGT\n", "invalid PAF record"),
+    ("empty_reads", "", "", ">a\nACGT\n", "empty sequences set"),
+    ("garbage_fasta", "not a fasta\n", "", ">a\nACGT\n", "invalid FASTA header"),
+])
+def test_malformed_inputs_fail_loudly(racon_cli, tmp_path, name, reads, ovl, tgt, msg):
+    r = tmp_path / "r.fasta"; r.write_text(reads)
+    o = tmp_path / "o.paf"; o.write_text(ovl)
+    t = tmp_path / "t.fasta"; t.write_text(tgt)
+    out = run_racon(racon_cli, [str(r), str(o), str(t)])
+    assert out.returncode != 0
+    assert msg in out.stderr
