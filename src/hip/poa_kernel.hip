@@ -39,7 +39,7 @@ constexpr int32_t kNegInf = -(1 << 28);
 constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
 constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
-constexpr uint32_t kMaxPre = 4;   // predecessor rows precomputed per row
+constexpr uint32_t kMaxPre = 2;   // predecessor rows precomputed per row
 
 // move byte encoding
 constexpr uint8_t kMvDiag = 0;
@@ -130,11 +130,6 @@ struct Shared {
   // cross-word extraction at the last chunk never reads out of bounds.
   uint64_t match[4][kMaxW / 64 + 1];
   uint64_t rd_block[64];  // row descriptors staged 64 at a time
-  // in/out degree mirrors: the serial graph phases (add_alignment, Kahn
-  // init, consensus walks) chase these per node; 4 KB keeps them at LDS
-  // latency without denting the VGPR-capped occupancy
-  uint8_t in_cnt[kMaxN];
-  uint8_t out_cnt[kMaxN];
   uint8_t seq[kMaxW];
 };
 
@@ -195,10 +190,10 @@ __device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared&, uint32
 }
 
 __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
-  uint32_t n_out = s.out_cnt[a];
+  uint32_t n_out = c.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
     if (out_edge_of(c, s, a, e) == b) {
-      uint32_t n_in = s.in_cnt[b];
+      uint32_t n_in = c.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
         if (c.in_edges[b * c.ME + f] == a) {
           c.in_weights[b * c.ME + f] += w;
@@ -208,16 +203,16 @@ __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t 
       return true;  // unreachable for a consistent graph
     }
   }
-  if (n_out >= c.ME || s.in_cnt[b] >= c.ME) {
+  if (n_out >= c.ME || c.in_cnt[b] >= c.ME) {
     c.status = kPoaEdgeOverflow;
     return false;
   }
   c.out_edges[a * c.ME + n_out] = static_cast<uint16_t>(b);
-  s.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
-  uint32_t n_in = s.in_cnt[b];
+  c.out_cnt[a] = static_cast<uint8_t>(n_out + 1);
+  uint32_t n_in = c.in_cnt[b];
   c.in_edges[b * c.ME + n_in] = static_cast<uint16_t>(a);
   c.in_weights[b * c.ME + n_in] = w;
-  s.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
+  c.in_cnt[b] = static_cast<uint8_t>(n_in + 1);
   return true;
 }
 
@@ -228,8 +223,8 @@ __device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
   }
   uint32_t id = c.num_nodes++;
   c.letters[id] = letter;
-  s.in_cnt[id] = 0;
-  s.out_cnt[id] = 0;
+  c.in_cnt[id] = 0;
+  c.out_cnt[id] = 0;
   c.ring_cnt[id] = 0;
   c.nseq[id] = 0;
   return static_cast<int32_t>(id);
@@ -358,7 +353,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
 __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
-    s.u.kahn.work[i] = s.in_cnt[i];
+    s.u.kahn.work[i] = c.in_cnt[i];
   }
   uint32_t qhead = 0, qtail = 0;
   for (uint32_t i = 0; i < n; ++i) {
@@ -368,7 +363,7 @@ __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
   }
   while (qhead < qtail) {
     uint16_t u = s.u.kahn.queue[qhead++];
-    uint32_t nout = s.out_cnt[u];
+    uint32_t nout = c.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
       uint16_t v = out_edge_of(c, s, u, e);
       if (--s.u.kahn.work[v] == 0) {
@@ -394,7 +389,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   uint32_t max_id = s.u.kahn.queue[0];
   for (uint32_t r = 0; r < n; ++r) {
     uint16_t nid = s.u.kahn.queue[r];
-    uint32_t nin = s.in_cnt[nid];
+    uint32_t nin = c.in_cnt[nid];
     for (uint32_t e = 0; e < nin; ++e) {
       uint16_t p = c.in_edges[nid * c.ME + e];
       int64_t w = c.in_weights[nid * c.ME + e];
@@ -418,7 +413,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
   // otherwise spin forever — those windows fail over to the CPU instead)
   uint32_t guard = 0;
   uint32_t prev_rank = 0;
-  while (s.out_cnt[max_id] != 0) {
+  while (c.out_cnt[max_id] != 0) {
     if (++guard > n || (guard > 1 && c.rank[max_id] <= prev_rank)) {
       c.status = kPoaConsensusOverflow;
       return -1;
@@ -426,10 +421,10 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
     prev_rank = c.rank[max_id];
     uint32_t rank0 = c.rank[max_id];
     // invalidate alternative branches
-    uint32_t nout = s.out_cnt[max_id];
+    uint32_t nout = c.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
       uint16_t endn = out_edge_of(c, s, max_id, e);
-      uint32_t nin = s.in_cnt[endn];
+      uint32_t nin = c.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
         uint16_t o = c.in_edges[endn * c.ME + f];
         if (o != max_id) {
@@ -443,7 +438,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
       uint16_t nid = s.u.kahn.queue[r];
       c.hb_score[nid] = -1;
       c.hb_pred[nid] = -1;
-      uint32_t nin = s.in_cnt[nid];
+      uint32_t nin = c.in_cnt[nid];
       for (uint32_t e = 0; e < nin; ++e) {
         uint16_t p = c.in_edges[nid * c.ME + e];
         if (c.hb_score[p] == -1) {
@@ -500,12 +495,12 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
   const uint32_t n = c.num_nodes;
   for (uint32_t node = lane; node < n; node += kLanes) {
     const uint32_t r = c.rank[node];
-    const uint8_t nin = s.in_cnt[node];
+    const uint8_t nin = c.in_cnt[node];
     uint16_t pred_row = 0;
     if (nin > 0) {
       pred_row = static_cast<uint16_t>(c.rank[c.in_edges[node * c.ME]] + 1);
     }
-    const uint32_t nout = s.out_cnt[node];
+    const uint32_t nout = c.out_cnt[node];
     uint8_t flags = (nout == 0) ? kRdEnd : 0;
     for (uint32_t e = 0; e < nout; ++e) {
       const uint32_t sr = c.rank[out_edge_of(c, s, node, e)];
@@ -521,7 +516,7 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
 // ---------- the mega-kernel ----------
 
 template <bool TIMED, uint32_t WB>
-__launch_bounds__(kLanes)
+__launch_bounds__(kLanes, 4)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                                   uint32_t num_windows) {
   if (blockIdx.x >= num_windows) {
@@ -530,7 +525,11 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
   const uint32_t win = window_base + blockIdx.x;
   const int lane = threadIdx.x;
   const PoaWindowDesc desc = a.windows[win];
-  const PoaLimits L = a.limits;
+  // the capacity model is fixed (PoaLimits defaults; PoaBatch never alters
+  // it) — compile-time limits keep slab addressing in immediates instead of
+  // ~100 live SGPRs that were spilling into v_readlane/writelane traffic in
+  // the DP row loop
+  constexpr PoaLimits L{};
   const uint32_t slab = desc.scratch_idx;
 
   __shared__ Shared s;
@@ -597,19 +596,19 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     c.rank[i] = static_cast<uint16_t>(i);
     uint8_t nin = 0;
     if (i == 0) {
-      s.in_cnt[i] = 0;
+      c.in_cnt[i] = 0;
     } else {
       nin = 1;
-      s.in_cnt[i] = 1;
+      c.in_cnt[i] = 1;
       c.in_edges[i * c.ME] = static_cast<uint16_t>(i - 1);
       c.in_weights[i * c.ME] = static_cast<int32_t>(bb_wts[i - 1]) + bb_wts[i];
     }
     uint8_t flags = 0;
     if (i + 1 < bb_len) {
-      s.out_cnt[i] = 1;
+      c.out_cnt[i] = 1;
       c.out_edges[i * c.ME] = static_cast<uint16_t>(i + 1);
     } else {
-      s.out_cnt[i] = 0;
+      c.out_cnt[i] = 0;
       flags = kRdEnd;
     }
     c.row_desc[i] = pack_rd(bb_seq[i], nin, static_cast<uint16_t>(i),
