@@ -17,6 +17,9 @@ namespace rga::hip {
 namespace {
 constexpr uint32_t kLanes = 64;
 
+// Band blocks for a requested width. The polisher's auto heuristic clamps
+// to 256 (K=4; see hip_polisher.cpp) — explicit --cudaaligner-band-width
+// values still select up to K=16.
 uint32_t pick_band_k(uint32_t band_width) {
   if (band_width == 0) return 8;  // default band 512
   uint32_t blocks = (band_width + 63) / 64;
@@ -25,11 +28,6 @@ uint32_t pick_band_k(uint32_t band_width) {
   return 16;
 }
 
-// The reference's auto heuristic (10% of mean overlap length,
-// cudapolisher.cpp:159-163) is tuned for its approximate banded DP; the
-// Myers band here is exact within the band and indel drift on long reads is
-// a short random walk, so clamp the auto choice to [256, 512]. Explicit
-// --cudaaligner-band-width values above 512 still select K=16.
 
 }  // namespace
 

@@ -86,3 +86,36 @@ def test_ngs_mode_short_reads(racon, tmp_path):
     truth = open(s["reference"]).read().splitlines()[1]
     draft = open(s["layout"]).read().splitlines()[1]
     assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth)
+
+
+def test_multi_contig_polish(racon, tmp_path):
+    """Several target contigs: output order, per-contig tags and collection
+    must follow the input order (reference polisher.cpp:506-532)."""
+    from racon_amd import synth
+    import shutil
+
+    dirs = []
+    for k in range(3):
+        d = tmp_path / f"g{k}"
+        dirs.append(synth.make_sample(d, genome_bp=12000 + 3000 * k, coverage=15, seed=30 + k))
+    reads = tmp_path / "reads.fasta"
+    ovls = tmp_path / "ovl.paf"
+    tgts = tmp_path / "targets.fasta"
+    with open(reads, "w") as out:
+        for k, s in enumerate(dirs):
+            for line in open(s["reads"]):
+                out.write(line.replace(">read", f">g{k}read"))
+    with open(ovls, "w") as out:
+        for k, s in enumerate(dirs):
+            for line in open(s["overlaps"]):
+                out.write(line.replace("read", f"g{k}read").replace("draft0", f"draft{k}"))
+    with open(tgts, "w") as out:
+        for k, s in enumerate(dirs):
+            for line in open(s["layout"]):
+                out.write(line.replace(">draft0", f">draft{k}"))
+    out = racon.polish(str(reads), str(ovls), str(tgts), threads=4)
+    assert [n.split()[0] for n, _ in out] == ["draft0", "draft1", "draft2"]
+    for k, (name, seq) in enumerate(out):
+        truth = open(dirs[k]["reference"]).read().splitlines()[1]
+        draft = open(dirs[k]["layout"]).read().splitlines()[1]
+        assert racon.edit_distance(seq, truth) < racon.edit_distance(draft, truth) * 0.2
