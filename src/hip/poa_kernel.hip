@@ -835,26 +835,47 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 phi = min(len, (pkhi + 1) * kLanes);
                 p_has0 = (pklo == 0);
               }
+              // validity of the contiguous range [plo+1, phi] (plus col 0
+              // when the pred row computed it) hoisted into ONE bitmask per
+              // edge — per-element compare+select chains were ~10% of the
+              // kernel's VALU issue
+              const uint32_t lo_col = plo + 1;
+              const uint32_t lo_w = lo_col > cbase ? min(lo_col - cbase, WB + 1) : 0;
+              const uint32_t hi_w = (phi + 1 > cbase) ? min(phi + 1 - cbase, WB + 1) : 0;
+              uint32_t okmask = (hi_w > lo_w) ? ((1u << hi_w) - (1u << lo_w)) : 0u;
+              if (cbase == 0 && p_has0) {
+                okmask |= 1u;  // col 0 sits outside [plo+1, phi] by construction
+              }
               const uint32_t colmax = c.MW - 1;
               if (r + 1 - p < kRing) {
                 const uint32_t slot = p % kRing;
+                if (cbase + WB <= colmax) {
 #pragma unroll
-                for (uint32_t w = 0; w <= WB; ++w) {
-                  const uint32_t col = cbase + w;
-                  const int32_t val = s.u.ring[slot][min(col, colmax)];
-                  const bool ok =
-                      (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
-                  pv[w] = ok ? val : kNegInf;
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    const int32_t val = s.u.ring[slot][cbase + w];
+                    pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
+                  }
+                } else {
+#pragma unroll
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    const int32_t val = s.u.ring[slot][min(cbase + w, colmax)];
+                    pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
+                  }
                 }
               } else {
                 const int16_t* gsrc = c.matrix + static_cast<size_t>(p) * c.MW;
+                if (cbase + WB <= colmax) {
 #pragma unroll
-                for (uint32_t w = 0; w <= WB; ++w) {
-                  const uint32_t col = cbase + w;
-                  const int32_t val = gsrc[min(col, colmax)];
-                  const bool ok =
-                      (col == 0) ? p_has0 : (col > plo && col <= phi && col <= len);
-                  pv[w] = ok ? val : kNegInf;
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    const int32_t val = gsrc[cbase + w];
+                    pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
+                  }
+                } else {
+#pragma unroll
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    const int32_t val = gsrc[min(cbase + w, colmax)];
+                    pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
+                  }
                 }
               }
             }
