@@ -879,25 +879,24 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 }
               }
             }
+            // no nown guard: out-of-range pv entries are -inf and propagate
 #pragma unroll
             for (uint32_t w = 0; w < WB; ++w) {
-              if (w < nown) {
-                const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
-                bd[w] = max(bd[w], pv[w] + sub);
-                bu[w] = max(bu[w], pv[w + 1] + c.g);
-              }
+              const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
+              bd[w] = max(bd[w], pv[w] + sub);
+              bu[w] = max(bu[w], pv[w + 1] + c.g);
             }
           }
         }
 
-        // local inclusive u-space scan over own columns
+        // local inclusive u-space scan over own columns (out-of-range
+        // candidates are already -inf; no per-element guard needed)
         int32_t us[WB];
         int32_t run = kNegInf;
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
           const int32_t j = static_cast<int32_t>(cbase + 1 + w);
-          const int32_t v = max(bd[w], bu[w]);
-          const int32_t u = (w < nown) ? v - j * c.g : kNegInf;
+          const int32_t u = max(bd[w], bu[w]) - j * c.g;
           run = max(run, u);
           us[w] = run;
         }
