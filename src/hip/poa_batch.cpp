@@ -277,10 +277,9 @@ std::vector<bool> PoaBatch::generate(bool trim) {
       aw = std::min(aw, arena_.band_width + 64);
     }
     const uint32_t need = (aw + 63) / 64;
-    if (need <= 5) return 5u;
-    if (need <= 8) return 8u;
-    if (need == 9) return 9u;  // one pass for 513..576-column windows
-    return 8u;                 // 8-wide multi-pass covers any width
+    // 9-wide single-pass was tried and spills 44-64 B/lane of scratch,
+    // which is catastrophically slow — 8-wide multi-pass covers any width
+    return need <= 5 ? 5u : 8u;
   };
   auto cost = [&](uint32_t w) {
     const uint32_t first = h_layer_index_[w];
