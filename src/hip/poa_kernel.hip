@@ -1096,10 +1096,6 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
         hipLaunchKernelGGL((poa_window_kernel<true, 5>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
-      case 9:
-        hipLaunchKernelGGL((poa_window_kernel<true, 9>), grid, block, 0, st, arena,
-                           window_base, num_windows);
-        break;
       default:
         hipLaunchKernelGGL((poa_window_kernel<true, 8>), grid, block, 0, st, arena,
                            window_base, num_windows);
@@ -1109,10 +1105,6 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
     switch (wb) {
       case 5:
         hipLaunchKernelGGL((poa_window_kernel<false, 5>), grid, block, 0, st, arena,
-                           window_base, num_windows);
-        break;
-      case 9:
-        hipLaunchKernelGGL((poa_window_kernel<false, 9>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
       default:
