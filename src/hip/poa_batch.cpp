@@ -150,9 +150,9 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
 }
 
 PoaBatch::~PoaBatch() {
-  hipSetDevice(device_);
+  (void)hipSetDevice(device_);
   if (d_pool_ != nullptr) {
-    hipFree(d_pool_);
+    (void)hipFree(d_pool_);
   }
   for (void* p : {static_cast<void*>(h_seq_), static_cast<void*>(h_weight_),
                   static_cast<void*>(h_layer_ends_), static_cast<void*>(h_layer_index_),
@@ -160,11 +160,11 @@ PoaBatch::~PoaBatch() {
                   static_cast<void*>(h_coverage_), static_cast<void*>(h_consensus_len_),
                   static_cast<void*>(h_status_)}) {
     if (p != nullptr) {
-      hipHostFree(p);
+      (void)hipHostFree(p);
     }
   }
   if (stream_ != nullptr) {
-    hipStreamDestroy(static_cast<hipStream_t>(stream_));
+    (void)hipStreamDestroy(static_cast<hipStream_t>(stream_));
   }
 }
 

@@ -44,8 +44,7 @@ constexpr uint32_t kMaxPre = 2;   // predecessor rows precomputed per row
 // move byte encoding
 constexpr uint8_t kMvDiag = 0;
 constexpr uint8_t kMvUp = 1;
-constexpr uint8_t kMvLeft = 2;
-constexpr uint8_t kMvInvalid = 3;
+constexpr uint8_t kMvLeft = 2;  // (value 3 is reserved/invalid)
 
 // Single-wavefront LDS visibility: waits only the LDS (lgkm) counter and
 // stops compiler reordering. Unlike __syncthreads(), it does NOT drain the
@@ -216,7 +215,7 @@ __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t 
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, Shared& s, uint8_t letter) {
+__device__ inline int32_t add_node_d(WindowCtx& c, Shared& /*s*/, uint8_t letter) {
   if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
     c.status = kPoaNodeOverflow;
     return -1;
@@ -654,7 +653,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     __syncthreads();
 
     const uint32_t n = c.num_nodes;
-    const uint32_t width = len + 1;
     const uint32_t chunks = (len + kLanes - 1) / kLanes;
     // banded mode (-b): static band around the rank diagonal
     const bool banded = (c.bw != 0) && (c.bw < len);
