@@ -37,17 +37,15 @@ namespace {
 constexpr int kLanes = 64;
 constexpr int32_t kNegInf = -(1 << 28);
 constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
-// DP ring rows are stored in group-padded form: 8-column groups at an
-// element stride of 10. The lane-blocked access pattern (stride 8 x int16
-// across lanes) would otherwise hit every LDS bank 8 ways, and at full
-// residency the LDS unit saturates before the VALUs do; with stride-10
-// groups the per-lane group bases advance 5 banks apiece (5 coprime 32),
-// spreading a 64-lane access across all banks (2-way worst case). The map
-// is injective by construction (groups own disjoint 10-element slots).
-constexpr uint32_t kMaxWPad = (kMaxW / 8) * 10;
+// DP ring rows are padded and offset-swizzled: the lane-blocked access
+// pattern (stride 8 x int16 across lanes) would otherwise hit every LDS
+// bank 8 ways, and at full residency the LDS unit saturates before the
+// VALUs do. col -> col + 2*((col/8) % 32) spreads a 64-lane strided access
+// over all 32 banks (2-way worst case).
+constexpr uint32_t kMaxWPad = kMaxW + 64;
 
 __device__ inline uint32_t ring_swz(uint32_t col) {
-  return (col >> 3) * 10 + (col & 7);
+  return col + 2 * ((col >> 3) & 31);
 }
 constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
