@@ -37,16 +37,6 @@ namespace {
 constexpr int kLanes = 64;
 constexpr int32_t kNegInf = -(1 << 28);
 constexpr uint32_t kMaxW = 1024;  // LDS row width; matrix_width must fit
-// DP ring rows are padded and offset-swizzled: the lane-blocked access
-// pattern (stride 8 x int16 across lanes) would otherwise hit every LDS
-// bank 8 ways, and at full residency the LDS unit saturates before the
-// VALUs do. col -> col + 2*((col/8) % 32) spreads a 64-lane strided access
-// over all 32 banks (2-way worst case).
-constexpr uint32_t kMaxWPad = kMaxW + 64;
-
-__device__ inline uint32_t ring_swz(uint32_t col) {
-  return col + 2 * ((col >> 3) & 31);
-}
 constexpr uint32_t kMaxN = 2048;  // LDS graph mirrors; max_nodes must fit
 constexpr uint32_t kRing = 4;     // DP rows kept in LDS
 constexpr uint32_t kMaxPre = 2;   // predecessor rows precomputed per row
@@ -127,7 +117,7 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
 // stays in the global slabs, hidden by co-residency.
 struct Shared {
   union {
-    int16_t ring[kRing][kMaxWPad];  // DP rows (slot = row % kRing), swizzled
+    int16_t ring[kRing][kMaxW];  // DP rows (slot = row % kRing)
     struct {
       uint16_t work[kMaxN];   // Kahn in-degree scratch
       uint16_t queue[kMaxN];  // Kahn FIFO == topological order
@@ -733,7 +723,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           }
         }
         if (r + 1 - p < kRing) {
-          return s.u.ring[p % kRing][ring_swz(col)];
+          return s.u.ring[p % kRing][col];
         }
         return c.matrix[static_cast<size_t>(p) * c.MW + col];
       };
@@ -762,7 +752,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           if (store_row) {
             Hrow[0] = static_cast<int16_t>(h0);
           }
-          ring_row[ring_swz(0)] = static_cast<int16_t>(h0);
+          ring_row[0] = static_cast<int16_t>(h0);
           // column 0 is always a vertical chain through the argmax edge
           Mrow[0] = static_cast<uint8_t>(kMvUp | (e0 << 2));
         }
@@ -860,13 +850,13 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 if (cbase + WB <= colmax) {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
-                    const int32_t val = s.u.ring[slot][ring_swz(cbase + w)];
+                    const int32_t val = s.u.ring[slot][cbase + w];
                     pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
                   }
                 } else {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
-                    const int32_t val = s.u.ring[slot][ring_swz(min(cbase + w, colmax))];
+                    const int32_t val = s.u.ring[slot][min(cbase + w, colmax)];
                     pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
                   }
                 }
@@ -952,7 +942,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             if (store_row) {
               Hrow[j] = static_cast<int16_t>(h16);
             }
-            ring_row[ring_swz(j)] = static_cast<int16_t>(h16);
+            ring_row[j] = static_cast<int16_t>(h16);
             Mrow[j] = mv;
             if (j == len) {
               h_sel = h;
