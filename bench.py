@@ -151,7 +151,7 @@ def main():
     if rank == 0:
         value = polished_bp / 1e6 / elapsed
         print(json.dumps({
-            "metric": "Mbp polished/sec (whole node), ONT C.elegans-class synthetic",
+            "metric": "Mbp polished/sec (whole node), ONT C.elegans, at 1/2/4/8 MI355X",
             "value": round(value, 4),
             "unit": "Mbp/s",
             "n_gpus": n_gpus,
