@@ -49,6 +49,23 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
     exit(1);
   }
 
+  // The kernel compiles the capacity model as constants (poa_kernel.hip
+  // keeps slab addressing in immediates); this object must not deviate
+  // from the PoaLimits defaults without recompiling the kernel.
+  {
+    PoaLimits defaults;
+    if (limits_.max_seq_len != defaults.max_seq_len ||
+        limits_.max_nodes != defaults.max_nodes ||
+        limits_.max_edges != defaults.max_edges ||
+        limits_.max_ring != defaults.max_ring ||
+        limits_.matrix_width != defaults.matrix_width ||
+        limits_.max_consensus != defaults.max_consensus) {
+      fprintf(stderr, "[rga::hip::PoaBatch] error: PoaLimits diverged from the "
+                      "compile-time kernel capacity model!\n");
+      exit(1);
+    }
+  }
+
   RGA_HIP_CHECK(hipSetDevice(device_));
   hipStream_t s;
   RGA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
