@@ -12,3 +12,12 @@ PY
 ./build-asan/racon -t 4 /tmp/asan_sample/reads.fasta /tmp/asan_sample/overlaps.paf \
     /tmp/asan_sample/layout.fasta > /dev/null
 echo "asan OK"
+
+# deeper pass: the reference golden workload end-to-end under ASan
+if [ -d /root/reference/test/data ]; then
+  ./build-asan/racon -t 8 -m 5 -x -4 -g -8 \
+      /root/reference/test/data/sample_reads.fastq.gz \
+      /root/reference/test/data/sample_overlaps.paf.gz \
+      /root/reference/test/data/sample_layout.fasta.gz > /dev/null
+  echo "asan golden OK"
+fi
