@@ -233,6 +233,21 @@ class HipPolisher : public Polisher {
       Polisher::polish(dst, drop_unpolished);
       return;
     }
+    // int16 device scores bound the usable score magnitudes
+    // (poa_batch.cpp guard); prefer the CPU engine over dying
+    {
+      const int32_t worst_param =
+          std::max({std::abs(static_cast<int32_t>(config_.match)),
+                    std::abs(static_cast<int32_t>(config_.mismatch)),
+                    std::abs(static_cast<int32_t>(config_.gap))});
+      if (static_cast<int32_t>(2047 + 1024) * worst_param > 32000) {
+        fprintf(stderr,
+                "[racon::HipPolisher] warning: score parameters too large for the "
+                "int16 GPU POA scores; polishing windows on the CPU instead\n");
+        Polisher::polish(dst, drop_unpolished);
+        return;
+      }
+    }
 
     logger_->log();
 
