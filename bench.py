@@ -111,12 +111,12 @@ def main():
             maxlen = int(sizes.max().item())
             padded = torch.zeros(maxlen, dtype=torch.uint8, device=device)
             padded[: t.numel()] = t
-            if rank == 0:
-                bufs = [torch.empty(maxlen, dtype=torch.uint8, device=device)
-                        for _ in range(world)]
-                dist.gather(padded, bufs, dst=0)
-            else:
-                dist.gather(padded, None, dst=0)
+            # all_gather rather than gather: gather is not supported on the
+            # NCCL (RCCL) backend; the padded all_gather is one extra hop of
+            # xGMI traffic and works on both backends
+            bufs = [torch.empty(maxlen, dtype=torch.uint8, device=device)
+                    for _ in range(world)]
+            dist.all_gather(bufs, padded)
         return sum(len(s) for _, s in out)
 
     def barrier_sync():
