@@ -74,7 +74,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   const size_t per_window = slab_bytes(limits_) + kSeqArenaPerWindow +
                             limits_.max_consensus * 3 + sizeof(PoaWindowDesc) + 1024;
   num_slabs_ = static_cast<uint32_t>(
-      std::min<size_t>(2048, std::max<size_t>(32, mem_budget / per_window)));
+      std::min<size_t>(4096, std::max<size_t>(32, mem_budget / per_window)));
   seq_arena_cap_ = static_cast<size_t>(num_slabs_) * kSeqArenaPerWindow;
 
   const size_t max_layers = static_cast<size_t>(num_slabs_) * (max_depth_ + 1);
