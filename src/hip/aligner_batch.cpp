@@ -41,8 +41,10 @@ AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
 
   // arena split: the per-column band state dominates — per alignment
   // ~ (m+1) * K * (16 B Pv/Mv + 4 B S); seqs/path/peq are q+t-scale.
-  // Cap the device pool so construction stays cheap even on 288 GB parts.
-  const size_t pool = std::min<size_t>(mem_budget, 24ull << 30);
+  // Cap the device pool so construction stays cheap and the POA arenas
+  // (pooled at the same time) keep headroom on 288 GB parts: at K=4 a
+  // 20 kbp alignment's state is ~1.3 MB, so 12 GB holds ~9k alignments.
+  const size_t pool = std::min<size_t>(mem_budget, 12ull << 30);
   seq_cap_ = std::max<size_t>(16u << 20, pool / 96);
   path_cap_ = seq_cap_;
   peq_cap_u64_ = seq_cap_ / 8;  // 4 codes per 64 bases = q_bytes/2 of u64s is
