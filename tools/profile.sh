@@ -7,6 +7,7 @@
 #   tools/profile.sh trace [bench args...]   # kernel trace + per-kernel stats
 #   tools/profile.sh pmc   [bench args...]   # SQ wait/issue counters
 #   tools/profile.sh hbm   [bench args...]   # L2 fetch/write bytes
+#   tools/profile.sh lds   [bench args...]   # LDS issue/stall counters
 # Output lands in profiles/<mode>/ (CSV + summary).
 set -euo pipefail
 cd "$(dirname "$0")/.."
@@ -31,8 +32,12 @@ case "$MODE" in
     (cd /tmp && rocprofv3 --pmc TCC_EA0_RDREQ_sum TCC_EA0_WRREQ_sum --stats \
         -d "$OLDPWD/$OUT" -- "${BENCH[@]}")
     ;;
+  lds)
+    (cd /tmp && rocprofv3 --pmc SQ_INSTS_LDS SQ_WAIT_INST_LDS SQ_WAVE_CYCLES SQ_WAVES \
+        --stats -d "$OLDPWD/$OUT" -- "${BENCH[@]}")
+    ;;
   *)
-    echo "unknown mode: $MODE (trace|pmc|hbm)" >&2
+    echo "unknown mode: $MODE (trace|pmc|hbm|lds)" >&2
     exit 1
     ;;
 esac
