@@ -25,10 +25,12 @@ REPO = Path(__file__).resolve().parent
 sys.path.insert(0, str(REPO / "build"))
 sys.path.insert(0, str(REPO))
 
-# One process per GPU: restrict HIP to this rank's device before any HIP init.
+# One process per GPU: restrict HIP to this rank's device before any HIP
+# init — ALWAYS, also for single-rank runs, so an N=1 measurement on an
+# 8-GPU node does not silently spread its in-process batches over all
+# devices and misreport single-GPU throughput.
 LOCAL_RANK = int(os.environ.get("LOCAL_RANK", 0))
-if "WORLD_SIZE" in os.environ and int(os.environ["WORLD_SIZE"]) > 1:
-    os.environ.setdefault("HIP_VISIBLE_DEVICES", str(LOCAL_RANK))
+os.environ.setdefault("HIP_VISIBLE_DEVICES", str(LOCAL_RANK))
 
 import torch  # noqa: E402
 import torch.distributed as dist  # noqa: E402
