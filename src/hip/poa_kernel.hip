@@ -754,7 +754,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           }
           ring_row[0] = static_cast<int16_t>(h0);
           // column 0 is always a vertical chain through the argmax edge
-          Mrow[0] = static_cast<uint8_t>(kMvUp | (e0 << 2));
+          // (moves layout is shifted: col j lives at j-1, col 0 at MW-1,
+          // so each lane's 8 move bytes form one aligned u64 store)
+          Mrow[c.MW - 1] = static_cast<uint8_t>(kMvUp | (e0 << 2));
         }
       }
 
@@ -908,6 +910,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
 
         // finalize own columns: h, moves, stores
         int32_t h_sel = kNegInf;
+        uint64_t mvpack = 0;  // 8 move bytes -> one aligned store at Mrow[cbase]
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
           if (w < nown) {
@@ -943,11 +946,21 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               Hrow[j] = static_cast<int16_t>(h16);
             }
             ring_row[j] = static_cast<int16_t>(h16);
-            Mrow[j] = mv;
+            if (WB == 8) {
+              mvpack |= static_cast<uint64_t>(mv) << (8 * w);
+            } else {
+              Mrow[j - 1] = mv;  // shifted layout, per-byte for odd widths
+            }
             if (j == len) {
               h_sel = h;
             }
           }
+        }
+        if (WB == 8 && cbase < len) {
+          // one aligned u64 store covers the lane's 8 move bytes
+          // (cbase is a multiple of 8 exactly when WB == 8); bytes beyond
+          // nown encode columns past the row end and are never read back
+          *reinterpret_cast<uint64_t*>(Mrow + cbase) = mvpack;
         }
         if (len > base && len <= base + kLanes * WB) {
           const int owner = static_cast<int>((len - 1 - base) / WB);
@@ -986,7 +999,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           rec_seq = static_cast<int32_t>(j - 1);
           prev_j = j - 1;
         } else {
-          const uint8_t mv = c.moves[static_cast<size_t>(i) * c.MW + j];
+          const uint8_t mv =
+              c.moves[static_cast<size_t>(i) * c.MW + (j != 0 ? j - 1 : c.MW - 1)];
           const uint8_t type = mv & 3;
           const uint32_t e = mv >> 2;
           const uint64_t rd = c.row_desc[i - 1];
