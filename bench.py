@@ -44,7 +44,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
-    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--genome-mbp", type=float, default=None,
                     help="genome Mbp per GPU shard (default 12.5 = C.elegans/8)")
     ap.add_argument("--coverage", type=int, default=30)
