@@ -1,5 +1,11 @@
 // _racon: python bindings over the native engine, used by tests/ and bench.py.
 // The compute path is pure C++/HIP; python only drives files in, FASTA out.
+//
+// Error semantics: fatal input errors (bad extensions, malformed records,
+// missing GPU when requested) terminate the process with a message on
+// stderr — the same die-on-error contract the reference pins with its
+// EXPECT_DEATH tests (racon_test.cpp:55-86). Callers that need isolation
+// run polish in a subprocess (as the CLI tests do).
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
