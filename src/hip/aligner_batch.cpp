@@ -129,11 +129,14 @@ int32_t AlignerBatch::reserve_span(const char* q, uint32_t q_len, const char* t,
   }
   const size_t bytes = static_cast<size_t>(q_len) + t_len;
   // per-lane share of the wave's tb region (the wave total is 64 of these,
-  // and the arena-capacity check is against the sum of all shares); allow
-  // ~2 sub-launches' worth per fill
+  // and the arena-capacity check is against the sum of all shares). The
+  // per-fill cap is HALF a tb arena: a greedy first fill at the full cap
+  // would swallow most of the overlap queue and starve the other batch
+  // threads (the pull model balances only when rounds are smaller than
+  // queue/threads).
   const uint64_t tb_need = static_cast<uint64_t>(t_len + 1) * band_k_ * 2;
   if (overlaps_.size() >= max_alignments_ || seq_bytes_ + bytes > seq_cap_ ||
-      path_bytes_ + bytes > path_cap_ || tb_reserved_ + tb_need > 2 * tb_cap_u64_) {
+      path_bytes_ + bytes > path_cap_ || tb_reserved_ + tb_need > tb_cap_u64_ / 2) {
     return -1;
   }
 
