@@ -101,6 +101,7 @@ AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
   d_waves_ = reinterpret_cast<AlnWaveDesc*>(base + o_waves);
   arena_.order = d_order_;
   arena_.waves = d_waves_;
+  arena_.lanes_per_wave = kLanes;
   arena_.limits = limits_;
 }
 
@@ -208,18 +209,28 @@ void AlignerBatch::run() {
   std::copy(order.begin(), order.end(), h_order_);
   RGA_HIP_CHECK(hipMemcpyAsync(d_order_, h_order_, na * 4, hipMemcpyHostToDevice, s));
 
-  // greedy sub-launches bounded by the tb/s/peq arenas
+  // greedy sub-launches bounded by the tb/s/peq arenas. Waves are
+  // deliberately under-filled for small jobs: at 64 alignments/wave a
+  // typical polish run yields ~1.5 waves per CU and every gather latency
+  // lands on the wall clock; 16-32/wave gives each CU interleavable waves
+  // at the cost of idle lanes (which issue no extra instructions).
   const uint32_t K = band_k_;
-  uint32_t wave_begin = 0;
-  const uint32_t num_waves_total = (na + kLanes - 1) / kLanes;
+  uint32_t lanes = kLanes;
+  if (na < 16384) {
+    lanes = 16;
+  } else if (na < 49152) {
+    lanes = 32;
+  }
+  uint32_t wave_begin = 0;  // in wave units
+  const uint32_t num_waves_total = (na + lanes - 1) / lanes;
   while (wave_begin < num_waves_total) {
     uint64_t peq_off = 0, tb_off = 0, s_off = 0;
     uint32_t w = wave_begin;
     uint32_t launch_waves = 0;
     for (; w < num_waves_total; ++w) {
       uint32_t nb = K, mmax = 0;
-      for (uint32_t l = 0; l < kLanes; ++l) {
-        const uint32_t slot = w * kLanes + l;
+      for (uint32_t l = 0; l < lanes; ++l) {
+        const uint32_t slot = w * lanes + l;
         if (slot >= na) break;
         const AlnDesc& d = h_descs_[h_order_[slot]];
         nb = std::max(nb, (d.q_len + 63) / 64);
@@ -247,14 +258,15 @@ void AlignerBatch::run() {
     // a single wave exceeding the arena alone can't be helped; it still has
     // its own full region (offsets 0) — the caps guarantee this fits because
     // max_len * K * 2 * 64 * 8 is carved into tb_cap by construction.
-    const uint32_t launch_slots =
-        std::min(na - wave_begin * kLanes, launch_waves * kLanes);
+    const uint32_t launch_align =
+        std::min(na - wave_begin * lanes, launch_waves * lanes);
     RGA_HIP_CHECK(hipMemcpyAsync(d_waves_, h_waves_, launch_waves * sizeof(AlnWaveDesc),
                                  hipMemcpyHostToDevice, s));
     AlnDeviceArena launch_arena = arena_;
-    launch_arena.order = d_order_ + wave_begin * kLanes;
+    launch_arena.order = d_order_ + wave_begin * lanes;
     launch_arena.waves = d_waves_;
-    launch_aligner_kernel(launch_arena, launch_waves, launch_slots, K, stream_);
+    launch_arena.lanes_per_wave = lanes;
+    launch_aligner_kernel(launch_arena, launch_waves, launch_align, K, stream_);
     RGA_HIP_CHECK(hipStreamSynchronize(s));
     wave_begin += launch_waves;
   }

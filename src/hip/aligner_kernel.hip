@@ -53,11 +53,13 @@ __device__ inline int32_t score_at(int32_t S, uint64_t Pv, uint64_t Mv, uint32_t
 
 template <int K>
 __launch_bounds__(kLanes)
-__global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
+__global__ void myers_kernel(AlnDeviceArena a, uint32_t num_align) {
   const uint32_t wave = blockIdx.x;
   const int lane = threadIdx.x;
-  const uint32_t slot = wave * kLanes + lane;
-  const bool active = slot < num_slots;
+  // lanes_per_wave < 64 under-fills waves on purpose: small jobs otherwise
+  // give each CU ~1 wave and every load latency lands on the wall clock
+  const uint32_t slot = wave * a.lanes_per_wave + lane;
+  const bool active = lane < static_cast<int>(a.lanes_per_wave) && slot < num_align;
 
   const AlnWaveDesc wd = a.waves[wave];
   const uint32_t idx = active ? a.order[slot] : 0u;
@@ -274,19 +276,19 @@ __global__ void myers_kernel(AlnDeviceArena a, uint32_t num_slots) {
 
 }  // namespace
 
-void launch_aligner_kernel(const AlnDeviceArena& arena, uint32_t num_waves, uint32_t num_slots,
+void launch_aligner_kernel(const AlnDeviceArena& arena, uint32_t num_waves, uint32_t num_align,
                            uint32_t band_k, void* stream) {
   auto s = static_cast<hipStream_t>(stream);
   switch (band_k) {
     case 4:
-      hipLaunchKernelGGL(myers_kernel<4>, dim3(num_waves), dim3(kLanes), 0, s, arena, num_slots);
+      hipLaunchKernelGGL(myers_kernel<4>, dim3(num_waves), dim3(kLanes), 0, s, arena, num_align);
       break;
     case 8:
-      hipLaunchKernelGGL(myers_kernel<8>, dim3(num_waves), dim3(kLanes), 0, s, arena, num_slots);
+      hipLaunchKernelGGL(myers_kernel<8>, dim3(num_waves), dim3(kLanes), 0, s, arena, num_align);
       break;
     default:
       hipLaunchKernelGGL(myers_kernel<16>, dim3(num_waves), dim3(kLanes), 0, s, arena,
-                         num_slots);
+                         num_align);
       break;
   }
 }

@@ -58,12 +58,15 @@ struct AlnDeviceArena {
   uint32_t* path_len;        // per alignment (original index)
   int32_t* status;           // per alignment
   int32_t* edit_distance;    // per alignment (diagnostic)
+  uint32_t lanes_per_wave;   // alignments packed per 64-lane wave; fewer
+                             // than 64 when the job is too small to give
+                             // every CU multiple waves (latency hiding)
   AlnLimits limits;
 };
 
-// Launches the K-block Myers kernel for `num_slots` = num_waves*64 sorted
-// lane slots (slots >= num_alignments are idle). band_k must be 4, 8 or 16.
-void launch_aligner_kernel(const AlnDeviceArena& arena, uint32_t num_waves, uint32_t num_slots,
+// Launches the K-block Myers kernel over `num_align` sorted alignments
+// packed arena.lanes_per_wave to a wave. band_k must be 4, 8 or 16.
+void launch_aligner_kernel(const AlnDeviceArena& arena, uint32_t num_waves, uint32_t num_align,
                            uint32_t band_k, void* stream);
 
 }  // namespace rga::hip
