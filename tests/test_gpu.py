@@ -225,3 +225,18 @@ def test_gpu_stress_configs(racon, tmp_path_factory, fasta_reader):
     truth = list(fasta_reader(s2["reference"]).values())[0]
     draft = list(fasta_reader(s2["layout"]).values())[0]
     assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth)
+
+
+def test_gpu_aligner_sub_launch_split(racon):
+    """Many large K=16 alignments exceed one traceback arena: the greedy
+    sub-launch splitter must produce identical numerics across launches."""
+    import random
+    rng = random.Random(11)
+    t = "".join(rng.choice("ACGT") for _ in range(15000))
+    pairs = [(_mutate(t, rng, 0.02, 0.02, 0.02), t) for _ in range(300)]
+    res = racon.gpu_align(pairs, band_width=1024)
+    ok = sum(1 for _, ed, st in res if st == 0)
+    assert ok >= 295, ok
+    for (q, tt), (cigar, ed, st) in list(zip(pairs, res))[:20]:
+        if st == 0:
+            assert ed == racon.edit_distance(q, tt)
