@@ -115,3 +115,25 @@ def test_preprocess_uniquifies_pairs(tmp_path):
     assert lines[0] == "@p11"
     assert lines[4] == "@p12"
     assert lines[8] == "@p21"
+
+
+def test_rampler_edge_cases(tmp_path, rampler):
+    # empty input: subsample errors cleanly; split writes nothing
+    empty = tmp_path / "empty.fasta"
+    empty.write_text("")
+    r = subprocess.run([rampler, "-o", str(tmp_path), "subsample", str(empty), "1000", "5"],
+                       capture_output=True, text=True)
+    assert r.returncode != 0
+    r = subprocess.run([rampler, "-o", str(tmp_path), "split", str(empty), "1000"],
+                       capture_output=True, text=True)
+    assert r.returncode == 0
+    assert "wrote 0 chunks" in r.stderr
+
+    # single record larger than the chunk size still lands in one chunk
+    single = tmp_path / "single.fasta"
+    write_fasta(single, [("big", "ACGT" * 1000)])
+    r = subprocess.run([rampler, "-o", str(tmp_path), "split", str(single), "100"],
+                       capture_output=True, text=True)
+    assert r.returncode == 0
+    assert (tmp_path / "single_0.fasta").exists()
+    assert not (tmp_path / "single_1.fasta").exists()
