@@ -114,34 +114,23 @@ std::unique_ptr<Overlap> Overlap::from_sam(std::string q_name, uint32_t flag, st
   return o;
 }
 
-namespace {
-
-template <typename K>
-bool lookup(const std::unordered_map<K, uint64_t>& map, const K& key, uint64_t* id) {
-  auto it = map.find(key);
-  if (it == map.end()) {
-    return false;
-  }
-  *id = it->second;
-  return true;
-}
-
-}  // namespace
-
 void Overlap::resolve_ids(const std::vector<std::unique_ptr<Sequence>>& sequences,
-                          const std::unordered_map<std::string, uint64_t>& name_to_id,
-                          const std::unordered_map<uint64_t, uint64_t>& id_to_id) {
+                          const SequenceIndex& index) {
   if (!is_valid_ || is_resolved_) {
     return;
   }
 
   if (!q_name_.empty()) {
-    if (!lookup(name_to_id, q_name_ + "q", &q_id_)) {
+    auto it = index.read_names.find(q_name_);
+    if (it == index.read_names.end()) {
       is_valid_ = false;
       return;
     }
+    q_id_ = it->second;
     std::string().swap(q_name_);
-  } else if (!lookup(id_to_id, q_id_ << 1 | 0, &q_id_)) {
+  } else if (q_id_ < index.read_ids.size()) {
+    q_id_ = index.read_ids[q_id_];
+  } else {
     is_valid_ = false;
     return;
   }
@@ -155,12 +144,16 @@ void Overlap::resolve_ids(const std::vector<std::unique_ptr<Sequence>>& sequence
   }
 
   if (!t_name_.empty()) {
-    if (!lookup(name_to_id, t_name_ + "t", &t_id_)) {
+    auto it = index.target_names.find(t_name_);
+    if (it == index.target_names.end()) {
       is_valid_ = false;
       return;
     }
+    t_id_ = it->second;
     std::string().swap(t_name_);
-  } else if (!lookup(id_to_id, t_id_ << 1 | 1, &t_id_)) {
+  } else if (t_id_ < index.target_ids.size()) {
+    t_id_ = index.target_ids[t_id_];
+  } else {
     is_valid_ = false;
     return;
   }

@@ -20,6 +20,16 @@ namespace rga {
 
 class Sequence;
 
+// Lookup tables mapping overlap-file identifiers (names for PAF/SAM, file
+// ordinals for MHAP) to global sequence indices. Reads whose name duplicates
+// a target share the target's slot — the polisher dedups them on load.
+struct SequenceIndex {
+  std::unordered_map<std::string, uint64_t> read_names;
+  std::unordered_map<std::string, uint64_t> target_names;
+  std::vector<uint64_t> read_ids;    // file-order read ordinal -> global index
+  std::vector<uint64_t> target_ids;  // file-order target ordinal -> global index
+};
+
 class Overlap {
  public:
   // MHAP record fields.
@@ -58,8 +68,7 @@ class Overlap {
   // Maps names / file-local ids to global sequence indices; marks the overlap
   // invalid when a name/id is unknown; validates lengths against sequences.
   void resolve_ids(const std::vector<std::unique_ptr<Sequence>>& sequences,
-                   const std::unordered_map<std::string, uint64_t>& name_to_id,
-                   const std::unordered_map<uint64_t, uint64_t>& id_to_id);
+                   const SequenceIndex& index);
 
   // Aligns q vs t spans when no CIGAR is present (CPU Myers NW), then walks the
   // CIGAR to produce per-window breaking points; frees the CIGAR afterwards.

@@ -8,43 +8,35 @@
 
 namespace rga {
 
-constexpr uint64_t kChunkSize = 1024ull * 1024 * 1024;  // 1 GiB parse chunks
+namespace {
 
-// Compacts nullptr gaps out of src starting at begin; returns removed count.
+// Parse granularity for the streamed read/overlap passes; bounds peak RSS on
+// genome-scale inputs (capability parity: reference streams 1 GiB chunks,
+// src/polisher.cpp:26).
+constexpr uint64_t kParseChunk = 1ull << 30;
+
+// Stable in-place compaction of released (null) entries from position `from`
+// on; returns how many were dropped.
 template <class T>
-static uint64_t compact(std::vector<std::unique_ptr<T>>& src, uint64_t begin) {
-  uint64_t i = begin;
-  for (uint64_t j = begin; i < src.size(); ++i) {
-    if (src[i] != nullptr) {
-      continue;
-    }
-    j = std::max(j, i);
-    while (j < src.size() && src[j] == nullptr) {
-      ++j;
-    }
-    if (j >= src.size()) {
-      break;
-    }
-    if (i != j) {
-      src[i].swap(src[j]);
-    }
-  }
-  uint64_t removed = src.size() - i;
-  if (i < src.size()) {
-    src.resize(i);
-  }
-  return removed;
+uint64_t drop_nulls(std::vector<std::unique_ptr<T>>* v, uint64_t from) {
+  auto live_end = std::remove_if(v->begin() + from, v->end(),
+                                 [](const std::unique_ptr<T>& p) { return p == nullptr; });
+  const uint64_t dropped = static_cast<uint64_t>(v->end() - live_end);
+  v->erase(live_end, v->end());
+  return dropped;
 }
+
+}  // namespace
 
 std::unique_ptr<Polisher> createPolisher(const std::string& sequences_path,
                                          const std::string& overlaps_path,
                                          const std::string& target_path, PolisherConfig config) {
   if (config.type != PolisherType::kC && config.type != PolisherType::kF) {
-    fprintf(stderr, "[racon::createPolisher] error: invalid polisher type!\n");
+    fprintf(stderr, "[rga::createPolisher] error: invalid polisher type!\n");
     exit(1);
   }
   if (config.window_length == 0) {
-    fprintf(stderr, "[racon::createPolisher] error: invalid window length!\n");
+    fprintf(stderr, "[rga::createPolisher] error: invalid window length!\n");
     exit(1);
   }
 
@@ -72,6 +64,7 @@ Polisher::Polisher(std::unique_ptr<SequenceParser> sparser, std::unique_ptr<Over
       dummy_quality_(config.window_length, '!'),
       thread_pool_(std::make_unique<ThreadPool>(config.num_threads)),
       logger_(std::make_unique<Logger>()) {
+  // one POA engine per pool thread (engines hold per-alignment scratch)
   for (uint32_t i = 0; i < thread_pool_->num_threads(); ++i) {
     engines_.emplace_back(
         std::make_unique<poa::NWEngine>(config.match, config.mismatch, config.gap));
@@ -80,264 +73,291 @@ Polisher::Polisher(std::unique_ptr<SequenceParser> sparser, std::unique_ptr<Over
 
 Polisher::~Polisher() {
   if (logger_) {
-    logger_->total("[racon::Polisher::] total =");
+    logger_->total("[rga::Polisher] total =");
   }
 }
 
-void Polisher::initialize() {
-  if (!windows_.empty()) {
-    fprintf(stderr, "[racon::Polisher::initialize] warning: object already initialized!\n");
-    return;
-  }
-
-  logger_->log();
-
+// Targets are loaded whole (they are the windows' backbones and must stay
+// resident); each gets a slot in both the name table and the ordinal table.
+uint64_t Polisher::load_targets(SequenceIndex* index) {
   tparser_->reset();
   tparser_->parse(sequences_, static_cast<uint64_t>(-1));
 
-  uint64_t targets_size = sequences_.size();
-  if (targets_size == 0) {
-    fprintf(stderr, "[racon::Polisher::initialize] error: empty target sequences set!\n");
+  const uint64_t num_targets = sequences_.size();
+  if (num_targets == 0) {
+    fprintf(stderr, "[rga::Polisher] error: empty target sequences set!\n");
     exit(1);
   }
-
-  std::unordered_map<std::string, uint64_t> name_to_id;
-  std::unordered_map<uint64_t, uint64_t> id_to_id;
-  for (uint64_t i = 0; i < targets_size; ++i) {
-    name_to_id[sequences_[i]->name() + "t"] = i;
-    id_to_id[i << 1 | 1] = i;
+  index->target_ids.reserve(num_targets);
+  for (uint64_t t = 0; t < num_targets; ++t) {
+    index->target_names.emplace(sequences_[t]->name(), t);
+    index->target_ids.push_back(t);
   }
+  return num_targets;
+}
 
-  std::vector<bool> has_name(targets_size, true);
-  std::vector<bool> has_data(targets_size, true);
-  std::vector<bool> has_reverse_data(targets_size, false);
-
-  logger_->log("[racon::Polisher::initialize] loaded target sequences");
-  logger_->log();
-
-  uint64_t sequences_size = 0, total_sequences_length = 0;
+// Reads stream in kParseChunk pieces. A read whose name duplicates a target
+// (self-polishing inputs) must share the target's global slot: its bytes are
+// checked for equality-of-shape, its index entries alias the target, and the
+// duplicate Sequence is dropped before the next chunk arrives.
+void Polisher::load_reads(SequenceIndex* index, uint64_t num_targets,
+                          std::vector<bool>* keep_name, std::vector<bool>* keep_fwd,
+                          std::vector<bool>* keep_rev) {
+  uint64_t num_reads = 0;
+  uint64_t total_read_bases = 0;
 
   sparser_->reset();
-  while (true) {
-    uint64_t l = sequences_.size();
-    bool more = sparser_->parse(sequences_, kChunkSize);
+  bool more = true;
+  while (more) {
+    const uint64_t chunk_begin = sequences_.size();
+    more = sparser_->parse(sequences_, kParseChunk);
 
-    uint64_t n = 0;
-    for (uint64_t i = l; i < sequences_.size(); ++i, ++sequences_size) {
-      total_sequences_length += sequences_[i]->data().size();
+    uint64_t dups_in_chunk = 0;
+    for (uint64_t i = chunk_begin; i < sequences_.size(); ++i, ++num_reads) {
+      const Sequence& read = *sequences_[i];
+      total_read_bases += read.data().size();
 
-      auto it = name_to_id.find(sequences_[i]->name() + "t");
-      if (it != name_to_id.end()) {
-        if (sequences_[i]->data().size() != sequences_[it->second]->data().size() ||
-            sequences_[i]->quality().size() != sequences_[it->second]->quality().size()) {
+      auto dup = index->target_names.find(read.name());
+      if (dup != index->target_names.end()) {
+        const Sequence& target = *sequences_[dup->second];
+        if (read.data().size() != target.data().size() ||
+            read.quality().size() != target.quality().size()) {
           fprintf(stderr,
-                  "[racon::Polisher::initialize] error: duplicate sequence %s with unequal data\n",
-                  sequences_[i]->name().c_str());
+                  "[rga::Polisher] error: duplicate sequence %s with unequal data\n",
+                  read.name().c_str());
           exit(1);
         }
-        name_to_id[sequences_[i]->name() + "q"] = it->second;
-        id_to_id[sequences_size << 1 | 0] = it->second;
-        sequences_[i].reset();
-        ++n;
+        index->read_names[read.name()] = dup->second;
+        index->read_ids.push_back(dup->second);
+        sequences_[i].reset();  // alias the target; drop the duplicate bytes
+        ++dups_in_chunk;
       } else {
-        name_to_id[sequences_[i]->name() + "q"] = i - n;
-        id_to_id[sequences_size << 1 | 0] = i - n;
+        const uint64_t slot = i - dups_in_chunk;  // position after this chunk's compaction
+        index->read_names[read.name()] = slot;
+        index->read_ids.push_back(slot);
       }
     }
 
-    compact(sequences_, l);
-    if (!more) {
-      break;
-    }
+    drop_nulls(&sequences_, chunk_begin);
   }
 
-  if (sequences_size == 0) {
-    fprintf(stderr, "[racon::Polisher::initialize] error: empty sequences set!\n");
+  if (num_reads == 0) {
+    fprintf(stderr, "[rga::Polisher] error: empty sequences set!\n");
     exit(1);
   }
 
-  has_name.resize(sequences_.size(), false);
-  has_data.resize(sequences_.size(), false);
-  has_reverse_data.resize(sequences_.size(), false);
-
-  window_type_ = static_cast<double>(total_sequences_length) / sequences_size <= 1000
+  // long-read vs short-read pipeline switch: mean read length over 1000
+  // selects TGS windows (coverage-trimmed consensus)
+  window_type_ = (static_cast<double>(total_read_bases) / num_reads <= 1000)
                      ? WindowType::kNGS
                      : WindowType::kTGS;
 
-  logger_->log("[racon::Polisher::initialize] loaded sequences");
-  logger_->log();
+  // retention bitmaps for the release pass: targets keep their name and
+  // forward bytes; reads keep nothing until an overlap claims a strand
+  keep_name->assign(sequences_.size(), false);
+  keep_fwd->assign(sequences_.size(), false);
+  keep_rev->assign(sequences_.size(), false);
+  for (uint64_t t = 0; t < num_targets; ++t) {
+    (*keep_name)[t] = true;
+    (*keep_fwd)[t] = true;
+  }
+}
 
-  std::vector<std::unique_ptr<Overlap>> overlaps;
-
-  auto remove_invalid_overlaps = [&](uint64_t begin, uint64_t end) {
+// Overlaps stream in chunks; records are assumed grouped by query (the
+// convention of minimap/mhap outputs). Each completed query group is
+// filtered: error-threshold and self-overlap drops always, and in contig
+// mode only the longest overlap of the query survives (ties: the later
+// record wins, matching the pinned goldens).
+void Polisher::load_overlaps(const SequenceIndex& index, std::vector<bool>* keep_fwd,
+                             std::vector<bool>* keep_rev,
+                             std::vector<std::unique_ptr<Overlap>>* overlaps) {
+  auto filter_query_group = [&](uint64_t begin, uint64_t end) {
+    uint64_t best = UINT64_MAX;  // surviving overlap in contig mode
     for (uint64_t i = begin; i < end; ++i) {
-      if (overlaps[i] == nullptr) {
+      auto& o = (*overlaps)[i];
+      if (o == nullptr) {
         continue;
       }
-      if (overlaps[i]->error() > config_.error_threshold ||
-          overlaps[i]->q_id() == overlaps[i]->t_id()) {
-        overlaps[i].reset();
+      if (o->error() > config_.error_threshold || o->q_id() == o->t_id()) {
+        o.reset();
         continue;
       }
-      if (config_.type == PolisherType::kC) {
-        // keep only the longest overlap per query
-        for (uint64_t j = i + 1; j < end; ++j) {
-          if (overlaps[j] == nullptr) {
-            continue;
-          }
-          if (overlaps[i]->length() > overlaps[j]->length()) {
-            overlaps[j].reset();
-          } else {
-            overlaps[i].reset();
-            break;
-          }
-        }
+      if (config_.type != PolisherType::kC) {
+        continue;  // fragment correction keeps every passing overlap
+      }
+      if (best == UINT64_MAX) {
+        best = i;
+      } else if (o->length() >= (*overlaps)[best]->length()) {
+        (*overlaps)[best].reset();
+        best = i;
+      } else {
+        o.reset();
       }
     }
   };
 
   oparser_->reset();
-  uint64_t l = 0;
-  while (true) {
-    bool more = oparser_->parse(overlaps, kChunkSize);
+  uint64_t resume = 0;  // first record of the (possibly unfinished) last group
+  bool more = true;
+  while (more) {
+    more = oparser_->parse(*overlaps, kParseChunk);
 
-    uint64_t c = l;
-    for (uint64_t i = l; i < overlaps.size(); ++i) {
-      overlaps[i]->resolve_ids(sequences_, name_to_id, id_to_id);
-      if (!overlaps[i]->is_valid()) {
-        overlaps[i].reset();
+    uint64_t group_begin = resume;
+    for (uint64_t i = resume; i < overlaps->size(); ++i) {
+      (*overlaps)[i]->resolve_ids(sequences_, index);
+      if (!(*overlaps)[i]->is_valid()) {
+        (*overlaps)[i].reset();
         continue;
       }
-      while (overlaps[c] == nullptr) {
-        ++c;
+      while ((*overlaps)[group_begin] == nullptr) {
+        ++group_begin;
       }
-      if (overlaps[c]->q_id() != overlaps[i]->q_id()) {
-        remove_invalid_overlaps(c, i);
-        c = i;
+      if ((*overlaps)[group_begin]->q_id() != (*overlaps)[i]->q_id()) {
+        filter_query_group(group_begin, i);
+        group_begin = i;
       }
     }
+    uint64_t settled = group_begin;  // records before this are fully filtered
     if (!more) {
-      remove_invalid_overlaps(c, overlaps.size());
-      c = overlaps.size();
+      filter_query_group(group_begin, overlaps->size());
+      settled = overlaps->size();
     }
 
-    for (uint64_t i = l; i < c; ++i) {
-      if (overlaps[i] == nullptr) {
-        continue;
-      }
-      if (overlaps[i]->strand()) {
-        has_reverse_data[overlaps[i]->q_id()] = true;
-      } else {
-        has_data[overlaps[i]->q_id()] = true;
+    // survivors pin the strand bytes their aligner pass will read
+    for (uint64_t i = resume; i < settled; ++i) {
+      const auto& o = (*overlaps)[i];
+      if (o != nullptr) {
+        auto& keep = o->strand() ? *keep_rev : *keep_fwd;
+        keep[o->q_id()] = true;
       }
     }
 
-    uint64_t removed = compact(overlaps, l);
-    l = c - removed;
-    if (!more) {
-      break;
-    }
+    const uint64_t dropped = drop_nulls(overlaps, resume);
+    // conservative resume point: nulls past `settled` also count into
+    // `dropped`, so this may land a little before the true group start —
+    // harmless, resolve_ids is a no-op on already-resolved records
+    resume = settled - std::min(settled, dropped);
   }
 
-  std::unordered_map<std::string, uint64_t>().swap(name_to_id);
-  std::unordered_map<uint64_t, uint64_t>().swap(id_to_id);
-
-  if (overlaps.empty()) {
-    fprintf(stderr, "[racon::Polisher::initialize] error: empty overlap set!\n");
+  if (overlaps->empty()) {
+    fprintf(stderr, "[rga::Polisher] error: empty overlap set!\n");
     exit(1);
   }
+}
 
-  logger_->log("[racon::Polisher::initialize] loaded overlaps");
+// Cut every target into window_length backbones. Windows are stored flat in
+// (target, rank) order; first_window_of_target_ gives each target's base.
+void Polisher::build_windows(uint64_t num_targets) {
+  first_window_of_target_.assign(num_targets + 1, 0);
+  for (uint64_t t = 0; t < num_targets; ++t) {
+    const std::string& backbone = sequences_[t]->data();
+    const std::string& qual = sequences_[t]->quality();
+    const uint32_t target_len = static_cast<uint32_t>(backbone.size());
+
+    uint32_t rank = 0;
+    for (uint32_t begin = 0; begin < target_len; begin += config_.window_length, ++rank) {
+      const uint32_t len = std::min(begin + config_.window_length, target_len) - begin;
+      const char* q = qual.empty() ? dummy_quality_.data() : &qual[begin];
+      windows_.push_back(createWindow(t, rank, window_type_, &backbone[begin], len, q, len));
+    }
+    first_window_of_target_[t + 1] = first_window_of_target_[t] + rank;
+  }
+}
+
+// Walk every overlap's breaking points and hand each aligned segment to its
+// window. Segments shorter than 2% of a window or below the mean-quality
+// threshold are dropped; overlaps are freed as they are consumed.
+void Polisher::route_layers(std::vector<std::unique_ptr<Overlap>>& overlaps,
+                            uint64_t num_targets) {
+  targets_coverages_.assign(num_targets, 0);
+
+  const double min_segment = 0.02 * config_.window_length;
+  for (auto& op : overlaps) {
+    const Overlap& o = *op;
+    ++targets_coverages_[o.t_id()];
+
+    const Sequence& read = *sequences_[o.q_id()];
+    const std::string& strand_data = o.strand() ? read.reverse_complement() : read.data();
+    const std::string& strand_qual = o.strand() ? read.reverse_quality() : read.quality();
+    const bool has_qual = !read.quality().empty() || !read.reverse_quality().empty();
+
+    const auto& anchors = o.breaking_points();  // (target_pos, query_pos) pairs
+    for (uint32_t a = 0; a + 1 < anchors.size(); a += 2) {
+      const uint32_t q_from = anchors[a].second;
+      const uint32_t q_to = anchors[a + 1].second;
+      if (q_to - q_from < min_segment) {
+        continue;
+      }
+
+      if (has_qual) {
+        uint64_t qsum = 0;
+        for (uint32_t k = q_from; k < q_to; ++k) {
+          qsum += static_cast<uint8_t>(strand_qual[k]) - 33;
+        }
+        if (static_cast<double>(qsum) / (q_to - q_from) < config_.quality_threshold) {
+          continue;
+        }
+      }
+
+      const uint32_t t_from = anchors[a].first;
+      const uint32_t rank = t_from / config_.window_length;
+      const uint32_t window_origin = rank * config_.window_length;
+      const char* seg_qual = strand_qual.empty() ? nullptr : &strand_qual[q_from];
+
+      windows_[first_window_of_target_[o.t_id()] + rank]->add_layer(
+          &strand_data[q_from], q_to - q_from, seg_qual,
+          seg_qual == nullptr ? 0 : q_to - q_from, t_from - window_origin,
+          anchors[a + 1].first - window_origin - 1);
+    }
+
+    op.reset();
+  }
+}
+
+void Polisher::initialize() {
+  if (!windows_.empty()) {
+    fprintf(stderr, "[rga::Polisher] warning: object already initialized!\n");
+    return;
+  }
+
   logger_->log();
+  SequenceIndex index;
+  const uint64_t num_targets = load_targets(&index);
+  logger_->log("[rga::Polisher] loaded target sequences");
 
+  logger_->log();
+  std::vector<bool> keep_name, keep_fwd, keep_rev;
+  load_reads(&index, num_targets, &keep_name, &keep_fwd, &keep_rev);
+  logger_->log("[rga::Polisher] loaded sequences");
+
+  logger_->log();
+  std::vector<std::unique_ptr<Overlap>> overlaps;
+  load_overlaps(index, &keep_fwd, &keep_rev, &overlaps);
+  index = SequenceIndex();  // release the lookup tables before alignment
+  logger_->log("[rga::Polisher] loaded overlaps");
+
+  logger_->log();
   {
-    std::vector<std::future<void>> futures;
+    // free every byte no surviving overlap needs; reverse complements are
+    // materialized here (on the pool) rather than lazily under the aligner
+    std::vector<std::future<void>> releases;
+    releases.reserve(sequences_.size());
     for (uint64_t i = 0; i < sequences_.size(); ++i) {
-      futures.emplace_back(thread_pool_->submit(
-          [&](uint64_t j) {
-            sequences_[j]->release(has_name[j], has_data[j], has_reverse_data[j]);
-          },
+      releases.emplace_back(thread_pool_->submit(
+          [&](uint64_t j) { sequences_[j]->release(keep_name[j], keep_fwd[j], keep_rev[j]); },
           i));
     }
-    for (const auto& f : futures) {
-      f.wait();
+    for (const auto& r : releases) {
+      r.wait();
     }
   }
 
   find_overlap_breaking_points(overlaps);
 
   logger_->log();
-
-  std::vector<uint64_t> id_to_first_window_id(targets_size + 1, 0);
-  for (uint64_t i = 0; i < targets_size; ++i) {
-    uint32_t k = 0;
-    for (uint32_t j = 0; j < sequences_[i]->data().size(); j += config_.window_length, ++k) {
-      uint32_t length = std::min(j + config_.window_length,
-                                 static_cast<uint32_t>(sequences_[i]->data().size())) -
-                        j;
-      windows_.emplace_back(createWindow(
-          i, k, window_type_, &(sequences_[i]->data()[j]), length,
-          sequences_[i]->quality().empty() ? &(dummy_quality_[0]) : &(sequences_[i]->quality()[j]),
-          length));
-    }
-    id_to_first_window_id[i + 1] = id_to_first_window_id[i] + k;
-  }
-
-  targets_coverages_.assign(targets_size, 0);
-
-  for (uint64_t i = 0; i < overlaps.size(); ++i) {
-    ++targets_coverages_[overlaps[i]->t_id()];
-
-    const auto& sequence = sequences_[overlaps[i]->q_id()];
-    const auto& breaking_points = overlaps[i]->breaking_points();
-
-    for (uint32_t j = 0; j < breaking_points.size(); j += 2) {
-      if (breaking_points[j + 1].second - breaking_points[j].second <
-          0.02 * config_.window_length) {
-        continue;
-      }
-
-      if (!sequence->quality().empty() || !sequence->reverse_quality().empty()) {
-        const auto& quality =
-            overlaps[i]->strand() ? sequence->reverse_quality() : sequence->quality();
-        double average_quality = 0;
-        for (uint32_t k = breaking_points[j].second; k < breaking_points[j + 1].second; ++k) {
-          average_quality += static_cast<uint32_t>(quality[k]) - 33;
-        }
-        average_quality /= breaking_points[j + 1].second - breaking_points[j].second;
-        if (average_quality < config_.quality_threshold) {
-          continue;
-        }
-      }
-
-      uint64_t window_id = id_to_first_window_id[overlaps[i]->t_id()] +
-                           breaking_points[j].first / config_.window_length;
-      uint32_t window_start =
-          (breaking_points[j].first / config_.window_length) * config_.window_length;
-
-      const char* data = overlaps[i]->strand()
-                             ? &(sequence->reverse_complement()[breaking_points[j].second])
-                             : &(sequence->data()[breaking_points[j].second]);
-      uint32_t data_length = breaking_points[j + 1].second - breaking_points[j].second;
-
-      const char* quality =
-          overlaps[i]->strand()
-              ? (sequence->reverse_quality().empty()
-                     ? nullptr
-                     : &(sequence->reverse_quality()[breaking_points[j].second]))
-              : (sequence->quality().empty() ? nullptr
-                                             : &(sequence->quality()[breaking_points[j].second]));
-      uint32_t quality_length = quality == nullptr ? 0 : data_length;
-
-      windows_[window_id]->add_layer(data, data_length, quality, quality_length,
-                                     breaking_points[j].first - window_start,
-                                     breaking_points[j + 1].first - window_start - 1);
-    }
-
-    overlaps[i].reset();
-  }
-
-  logger_->log("[racon::Polisher::initialize] transformed data into windows");
+  build_windows(num_targets);
+  route_layers(overlaps, num_targets);
+  logger_->log("[rga::Polisher] transformed data into windows");
 }
 
 void Polisher::find_overlap_breaking_points(std::vector<std::unique_ptr<Overlap>>& overlaps) {
@@ -349,17 +369,17 @@ void Polisher::find_overlap_breaking_points(std::vector<std::unique_ptr<Overlap>
         i));
   }
 
-  uint64_t logger_step = futures.size() / 20;
+  const uint64_t bar_step = futures.size() / 20;
   for (uint64_t i = 0; i < futures.size(); ++i) {
     futures[i].wait();
-    if (logger_step != 0 && (i + 1) % logger_step == 0 && (i + 1) / logger_step < 20) {
-      logger_->bar("[racon::Polisher::initialize] aligning overlaps");
+    if (bar_step != 0 && (i + 1) % bar_step == 0 && (i + 1) / bar_step < 20) {
+      logger_->bar("[rga::Polisher] aligning overlaps");
     }
   }
-  if (logger_step != 0) {
-    logger_->bar("[racon::Polisher::initialize] aligning overlaps");
+  if (bar_step != 0) {
+    logger_->bar("[rga::Polisher] aligning overlaps");
   } else {
-    logger_->log("[racon::Polisher::initialize] aligned overlaps");
+    logger_->log("[rga::Polisher] aligned overlaps");
   }
 }
 
@@ -373,9 +393,9 @@ void Polisher::generate_consensus_cpu(std::vector<bool>& polished,
     }
     futures.emplace_back(thread_pool_->submit(
         [&](uint64_t j) -> std::pair<uint64_t, bool> {
-          uint32_t tid = thread_pool_->this_thread_id();
+          const uint32_t tid = thread_pool_->this_thread_id();
           if (tid == ~0u) {
-            fprintf(stderr, "[racon::Polisher::polish] error: thread identifier not present!\n");
+            fprintf(stderr, "[rga::Polisher] error: worker outside the thread pool!\n");
             exit(1);
           }
           return {j, windows_[j]->generate_consensus(*engines_[tid], config_.trim)};
@@ -383,44 +403,54 @@ void Polisher::generate_consensus_cpu(std::vector<bool>& polished,
         i));
   }
 
-  uint64_t logger_step = futures.size() / 20;
+  const uint64_t bar_step = futures.size() / 20;
   for (uint64_t i = 0; i < futures.size(); ++i) {
     auto result = futures[i].get();
     polished[result.first] = result.second;
-    if (logger_step != 0 && (i + 1) % logger_step == 0 && (i + 1) / logger_step < 20) {
-      logger_->bar("[racon::Polisher::polish] generating consensus");
+    if (bar_step != 0 && (i + 1) % bar_step == 0 && (i + 1) / bar_step < 20) {
+      logger_->bar("[rga::Polisher] generating consensus");
     }
   }
-  if (logger_step != 0) {
-    logger_->bar("[racon::Polisher::polish] generating consensus");
+  if (bar_step != 0) {
+    logger_->bar("[rga::Polisher] generating consensus");
   } else {
-    logger_->log("[racon::Polisher::polish] generated consensus");
+    logger_->log("[rga::Polisher] generated consensus");
   }
 }
 
 void Polisher::collect(std::vector<std::unique_ptr<Sequence>>& dst, bool drop_unpolished,
                        const std::vector<bool>& polished) {
-  std::string polished_data;
-  uint32_t num_polished_windows = 0;
+  // windows are stored (target, rank)-ordered: concatenate consensuses until
+  // the next window restarts at rank 0, then emit the finished contig
+  std::string contig;
+  uint32_t contig_polished_windows = 0;
+
+  auto emit_contig = [&](uint64_t last_window) {
+    const Window& w = *windows_[last_window];
+    const double polished_ratio =
+        static_cast<double>(contig_polished_windows) / (w.rank() + 1);
+    if (!drop_unpolished || polished_ratio > 0) {
+      // output tags are part of the format contract (golden diffs read them):
+      // LN = length, RC = read count, XC = polished-window ratio; fragment
+      // correction prefixes "r"
+      std::string header = sequences_[w.id()]->name();
+      if (config_.type == PolisherType::kF) {
+        header += "r";
+      }
+      header += " LN:i:" + std::to_string(contig.size());
+      header += " RC:i:" + std::to_string(targets_coverages_[w.id()]);
+      header += " XC:f:" + std::to_string(polished_ratio);
+      dst.emplace_back(createSequence(header, contig));
+    }
+    contig.clear();
+    contig_polished_windows = 0;
+  };
 
   for (uint64_t i = 0; i < windows_.size(); ++i) {
-    num_polished_windows += polished[i] ? 1 : 0;
-    polished_data += windows_[i]->consensus();
-
-    if (i == windows_.size() - 1 || windows_[i + 1]->rank() == 0) {
-      double polished_ratio =
-          num_polished_windows / static_cast<double>(windows_[i]->rank() + 1);
-
-      if (!drop_unpolished || polished_ratio > 0) {
-        std::string tags = config_.type == PolisherType::kF ? "r" : "";
-        tags += " LN:i:" + std::to_string(polished_data.size());
-        tags += " RC:i:" + std::to_string(targets_coverages_[windows_[i]->id()]);
-        tags += " XC:f:" + std::to_string(polished_ratio);
-        dst.emplace_back(createSequence(sequences_[windows_[i]->id()]->name() + tags,
-                                        polished_data));
-      }
-      num_polished_windows = 0;
-      polished_data.clear();
+    contig_polished_windows += polished[i] ? 1 : 0;
+    contig += windows_[i]->consensus();
+    if (i + 1 == windows_.size() || windows_[i + 1]->rank() == 0) {
+      emit_contig(i);
     }
     windows_[i].reset();
   }

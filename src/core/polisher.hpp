@@ -83,6 +83,19 @@ class Polisher {
 
   std::unique_ptr<ThreadPool> thread_pool_;
   std::unique_ptr<Logger> logger_;
+
+ private:
+  // initialize() phases. Targets load whole; reads and overlaps stream in
+  // bounded chunks so peak RSS stays flat on genome-scale inputs.
+  uint64_t load_targets(SequenceIndex* index);
+  void load_reads(SequenceIndex* index, uint64_t num_targets, std::vector<bool>* keep_name,
+                  std::vector<bool>* keep_fwd, std::vector<bool>* keep_rev);
+  void load_overlaps(const SequenceIndex& index, std::vector<bool>* keep_fwd,
+                     std::vector<bool>* keep_rev, std::vector<std::unique_ptr<Overlap>>* overlaps);
+  void build_windows(uint64_t num_targets);
+  void route_layers(std::vector<std::unique_ptr<Overlap>>& overlaps, uint64_t num_targets);
+
+  std::vector<uint64_t> first_window_of_target_;
 };
 
 std::unique_ptr<Polisher> createPolisher(const std::string& sequences_path,

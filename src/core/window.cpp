@@ -8,88 +8,128 @@
 
 namespace rga {
 
+namespace {
+
+// Trim both consensus ends up to the first base whose coverage reaches the
+// floor. Returns false when the trimmed range is empty or inverted (the
+// window straddles a likely chimeric join and is left untrimmed).
+bool trim_to_coverage(std::string* consensus, const std::vector<uint32_t>& coverages,
+                      uint32_t floor) {
+  const int32_t n = static_cast<int32_t>(consensus->size());
+  int32_t lo = 0;
+  while (lo < n && coverages[lo] < floor) {
+    ++lo;
+  }
+  int32_t hi = n - 1;
+  while (hi >= 0 && coverages[hi] < floor) {
+    --hi;
+  }
+  if (lo >= hi) {
+    return false;
+  }
+  *consensus = consensus->substr(lo, hi - lo + 1);
+  return true;
+}
+
+}  // namespace
+
 std::shared_ptr<Window> createWindow(uint64_t id, uint32_t rank, WindowType type,
                                      const char* backbone, uint32_t backbone_length,
                                      const char* quality, uint32_t quality_length) {
   if (backbone_length == 0 || backbone_length != quality_length) {
-    fprintf(stderr, "[rga::createWindow] error: empty backbone sequence/unequal quality length!\n");
+    fprintf(stderr,
+            "[rga::createWindow] error: backbone must be non-empty with a "
+            "quality string of the same length\n");
     exit(1);
   }
   return std::make_shared<Window>(id, rank, type, backbone, backbone_length, quality,
                                   quality_length);
 }
 
-Window::Window(uint64_t id, uint32_t rank, WindowType type, const char* backbone,
-               uint32_t backbone_length, const char* quality, uint32_t quality_length)
-    : id_(id), rank_(rank), type_(type) {
-  sequences_.emplace_back(backbone, backbone_length);
-  qualities_.emplace_back(quality, quality_length);
-  positions_.emplace_back(0, 0);
+Window::Window(uint64_t contig_id, uint32_t rank, WindowType type, const char* backbone,
+               uint32_t backbone_len, const char* backbone_qual, uint32_t qual_len)
+    : contig_id_(contig_id), rank_(rank), type_(type) {
+  Layer base;
+  base.seq = backbone;
+  base.seq_len = backbone_len;
+  base.qual = backbone_qual;
+  base.qual_len = qual_len;
+  layers_.push_back(base);
 }
 
 void Window::add_layer(const char* sequence, uint32_t sequence_length, const char* quality,
                        uint32_t quality_length, uint32_t begin, uint32_t end) {
   if (sequence_length == 0 || begin == end) {
-    return;
+    return;  // nothing routed into this window
   }
   if (quality != nullptr && sequence_length != quality_length) {
-    fprintf(stderr, "[rga::Window::add_layer] error: unequal quality size!\n");
+    fprintf(stderr, "[rga::Window::add_layer] error: quality length does not match the segment\n");
     exit(1);
   }
-  if (begin >= end || begin > sequences_.front().second || end > sequences_.front().second) {
-    fprintf(stderr, "[rga::Window::add_layer] error: layer begin and end positions are invalid!\n");
+  const uint32_t backbone_len = backbone_length();
+  if (begin >= end || begin > backbone_len || end > backbone_len) {
+    fprintf(stderr, "[rga::Window::add_layer] error: segment span outside the backbone\n");
     exit(1);
   }
-  sequences_.emplace_back(sequence, sequence_length);
-  qualities_.emplace_back(quality, quality_length);
-  positions_.emplace_back(begin, end);
+  Layer l;
+  l.seq = sequence;
+  l.seq_len = sequence_length;
+  l.qual = quality;
+  l.qual_len = quality_length;
+  l.begin = begin;
+  l.end = end;
+  layers_.push_back(l);
 }
 
 std::vector<uint32_t> Window::layer_order() const {
-  std::vector<uint32_t> rank;
-  rank.reserve(sequences_.size());
-  for (uint32_t i = 0; i < sequences_.size(); ++i) {
-    rank.emplace_back(i);
+  std::vector<uint32_t> order(layers_.size());
+  for (uint32_t i = 0; i < order.size(); ++i) {
+    order[i] = i;
   }
-  // Unstable sort, matching the reference exactly (ref window.cpp:84-85):
-  // equal start positions keep libstdc++'s introsort order.
-  std::sort(rank.begin() + 1, rank.end(),
-            [&](uint32_t lhs, uint32_t rhs) { return positions_[lhs].first < positions_[rhs].first; });
-  return rank;
+  // Deliberately an UNSTABLE sort over everything but the backbone: the
+  // pinned CPU goldens encode libstdc++ introsort's ordering of equal start
+  // positions (reference behavior), so this must stay std::sort with a
+  // strict-weak begin-only comparison.
+  std::sort(order.begin() + 1, order.end(), [this](uint32_t a, uint32_t b) {
+    return layers_[a].begin < layers_[b].begin;
+  });
+  return order;
 }
 
 bool Window::generate_consensus(poa::NWEngine& engine, bool trim) {
-  if (sequences_.size() < 3) {
-    consensus_ = std::string(sequences_.front().first, sequences_.front().second);
+  const Layer& base = layers_.front();
+  if (layers_.size() < 3) {
+    // under-covered: pass the backbone through, flagged unpolished
+    consensus_.assign(base.seq, base.seq_len);
     return false;
   }
 
   poa::Graph graph;
-  graph.add_alignment(poa::Alignment(), sequences_.front().first, sequences_.front().second,
-                      qualities_.front().first, qualities_.front().second);
+  graph.add_alignment(poa::Alignment(), base.seq, base.seq_len, base.qual, base.qual_len);
 
-  std::vector<uint32_t> rank = layer_order();
+  // a layer within 1% of both window edges counts as spanning the whole
+  // backbone and is aligned against the full graph; anything shorter goes
+  // through the subgraph of its own backbone range
+  const uint32_t edge_margin = static_cast<uint32_t>(0.01 * base.seq_len);
+  const std::vector<uint32_t> order = layer_order();
+  for (uint32_t k = 1; k < order.size(); ++k) {
+    const Layer& l = layers_[order[k]];
 
-  uint32_t offset = static_cast<uint32_t>(0.01 * sequences_.front().second);
-  for (uint32_t j = 1; j < sequences_.size(); ++j) {
-    uint32_t i = rank[j];
-
-    poa::Alignment alignment;
-    if (positions_[i].first < offset &&
-        positions_[i].second > sequences_.front().second - offset) {
-      alignment = engine.align(sequences_[i].first, sequences_[i].second, graph);
+    poa::Alignment aln;
+    const bool spans_window = l.begin < edge_margin && l.end > base.seq_len - edge_margin;
+    if (spans_window) {
+      aln = engine.align(l.seq, l.seq_len, graph);
     } else {
-      std::vector<int32_t> mapping;
-      auto subgraph = graph.subgraph(positions_[i].first, positions_[i].second, &mapping);
-      alignment = engine.align(sequences_[i].first, sequences_[i].second, subgraph);
-      poa::Graph::update_alignment(&alignment, mapping);
+      std::vector<int32_t> node_map;
+      poa::Graph ranged = graph.subgraph(l.begin, l.end, &node_map);
+      aln = engine.align(l.seq, l.seq_len, ranged);
+      poa::Graph::update_alignment(&aln, node_map);
     }
 
-    if (qualities_[i].first == nullptr) {
-      graph.add_alignment(alignment, sequences_[i].first, sequences_[i].second);
+    if (l.qual == nullptr) {
+      graph.add_alignment(aln, l.seq, l.seq_len);
     } else {
-      graph.add_alignment(alignment, sequences_[i].first, sequences_[i].second,
-                          qualities_[i].first, qualities_[i].second);
+      graph.add_alignment(aln, l.seq, l.seq_len, l.qual, l.qual_len);
     }
   }
 
@@ -97,27 +137,12 @@ bool Window::generate_consensus(poa::NWEngine& engine, bool trim) {
   consensus_ = graph.generate_consensus(&coverages);
 
   if (type_ == WindowType::kTGS && trim) {
-    uint32_t average_coverage = static_cast<uint32_t>(sequences_.size() - 1) / 2;
-
-    int32_t begin = 0, end = static_cast<int32_t>(consensus_.size()) - 1;
-    for (; begin < static_cast<int32_t>(consensus_.size()); ++begin) {
-      if (coverages[begin] >= average_coverage) {
-        break;
-      }
-    }
-    for (; end >= 0; --end) {
-      if (coverages[end] >= average_coverage) {
-        break;
-      }
-    }
-
-    if (begin >= end) {
+    const uint32_t cov_floor = static_cast<uint32_t>(layers_.size() - 1) / 2;
+    if (!trim_to_coverage(&consensus_, coverages, cov_floor)) {
       fprintf(stderr,
-              "[rga::Window::generate_consensus] warning: "
-              "contig %lu might be chimeric in window %u!\n",
-              id_, rank_);
-    } else {
-      consensus_ = consensus_.substr(begin, end - begin + 1);
+              "[rga::Window::generate_consensus] warning: window %u of contig %lu "
+              "has no half-coverage core (possible chimera); left untrimmed\n",
+              rank_, static_cast<unsigned long>(contig_id_));
     }
   }
 

@@ -1,9 +1,12 @@
-// One POA consensus unit: a backbone slice of a target contig plus the read
-// segments (layers) routed into it. Behavioral parity with reference
-// src/window.{hpp,cpp}: <3 layers copies the backbone; layers are added in
-// start-position-sorted order (std::sort, same tie behavior); layers not
-// spanning ~98% of the window are aligned to a subgraph of the backbone range;
-// TGS windows trim consensus ends below half-average coverage.
+// One POA consensus task: a slice of a target contig (the backbone) plus the
+// read segments ("layers") that the breaking-point walk routed onto it.
+//
+// Capability parity with reference src/window.{hpp,cpp}: fewer than three
+// layers copies the backbone through unpolished; layers enter the graph in
+// start-position order (unstable sort — introsort tie order is part of the
+// pinned CPU goldens); a layer that does not reach within ~1% of both window
+// edges is aligned against a subgraph of its backbone range; long-read (kTGS)
+// windows have consensus ends below half-average coverage trimmed off.
 #pragma once
 
 #include <cstdint>
@@ -25,39 +28,56 @@ enum class WindowType {
 
 class Window {
  public:
-  Window(uint64_t id, uint32_t rank, WindowType type, const char* backbone,
-         uint32_t backbone_length, const char* quality, uint32_t quality_length);
+  // A sequence segment participating in this window's consensus. Views only:
+  // the bytes live in the Polisher's Sequence objects, which must outlive the
+  // window. Slot 0 is always the backbone (span unset).
+  struct Layer {
+    const char* seq = nullptr;
+    uint32_t seq_len = 0;
+    const char* qual = nullptr;  // nullptr when the source read had none
+    uint32_t qual_len = 0;
+    uint32_t begin = 0;  // backbone coordinates of the routed segment
+    uint32_t end = 0;
+  };
 
-  uint64_t id() const { return id_; }
+  Window(uint64_t contig_id, uint32_t rank, WindowType type, const char* backbone,
+         uint32_t backbone_len, const char* backbone_qual, uint32_t qual_len);
+
+  uint64_t id() const { return contig_id_; }
   uint32_t rank() const { return rank_; }
   WindowType type() const { return type_; }
   const std::string& consensus() const { return consensus_; }
   void set_consensus(std::string consensus) { consensus_ = std::move(consensus); }
 
-  uint32_t num_layers() const { return static_cast<uint32_t>(sequences_.size()); }
-  // Layer views (index 0 is the backbone).
-  const std::pair<const char*, uint32_t>& sequence(uint32_t i) const { return sequences_[i]; }
-  const std::pair<const char*, uint32_t>& quality(uint32_t i) const { return qualities_[i]; }
-  const std::pair<uint32_t, uint32_t>& position(uint32_t i) const { return positions_[i]; }
+  uint32_t num_layers() const { return static_cast<uint32_t>(layers_.size()); }
+  uint32_t backbone_length() const { return layers_.front().seq_len; }
 
-  // Layer order sorted by start position (backbone stays first); shared by the
-  // CPU and HIP consensus paths so results do not depend on the device.
+  // Pair-view accessors kept for the batch packers (index 0 = backbone).
+  std::pair<const char*, uint32_t> sequence(uint32_t i) const {
+    return {layers_[i].seq, layers_[i].seq_len};
+  }
+  std::pair<const char*, uint32_t> quality(uint32_t i) const {
+    return {layers_[i].qual, layers_[i].qual_len};
+  }
+
+  // Indices of all layers with the backbone first and the rest ordered by
+  // start position. Shared by the CPU and HIP consensus paths so a window's
+  // result never depends on which device polished it.
   std::vector<uint32_t> layer_order() const;
 
   void add_layer(const char* sequence, uint32_t sequence_length, const char* quality,
                  uint32_t quality_length, uint32_t begin, uint32_t end);
 
-  // CPU POA consensus; returns true when a real consensus was generated.
+  // CPU POA consensus; returns true when a real consensus was generated
+  // (false = backbone copy for under-covered windows).
   bool generate_consensus(poa::NWEngine& engine, bool trim);
 
  private:
-  uint64_t id_;
+  uint64_t contig_id_;
   uint32_t rank_;
   WindowType type_;
   std::string consensus_;
-  std::vector<std::pair<const char*, uint32_t>> sequences_;
-  std::vector<std::pair<const char*, uint32_t>> qualities_;
-  std::vector<std::pair<uint32_t, uint32_t>> positions_;
+  std::vector<Layer> layers_;
 };
 
 std::shared_ptr<Window> createWindow(uint64_t id, uint32_t rank, WindowType type,

@@ -223,22 +223,47 @@ void AlignerBatch::run() {
   }
   uint32_t wave_begin = 0;  // in wave units
   const uint32_t num_waves_total = (na + lanes - 1) / lanes;
+  std::vector<uint32_t> never_run;  // original indices of un-runnable waves
+  auto wave_needs = [&](uint32_t w, uint64_t* peq_need, uint64_t* tb_need,
+                        uint64_t* s_need) {
+    uint32_t nb = K, mmax = 0;
+    for (uint32_t l = 0; l < lanes; ++l) {
+      const uint32_t slot = w * lanes + l;
+      if (slot >= na) break;
+      const AlnDesc& d = h_descs_[h_order_[slot]];
+      nb = std::max(nb, (d.q_len + 63) / 64);
+      mmax = std::max(mmax, d.t_len);
+    }
+    *peq_need = static_cast<uint64_t>(nb) * 4 * kLanes;
+    *tb_need = static_cast<uint64_t>(mmax + 1) * K * 2 * kLanes;
+    *s_need = static_cast<uint64_t>(mmax + 1) * K * kLanes;
+    return std::make_pair(nb, mmax);
+  };
   while (wave_begin < num_waves_total) {
+    // a wave whose state alone exceeds an arena (giant target at a wide
+    // explicit band: e.g. 262 kbp at K=16 needs ~4.3 GB of tb) can never
+    // launch — fail its alignments to the CPU pairwise fallback instead of
+    // writing out of bounds. No construction-time carve guarantees a full
+    // max_len wave fits; this check is the guarantee.
+    {
+      uint64_t peq_need, tb_need, s_need;
+      wave_needs(wave_begin, &peq_need, &tb_need, &s_need);
+      if (peq_need > peq_cap_u64_ || tb_need > tb_cap_u64_ || s_need > s_cap_i32_) {
+        for (uint32_t l = 0; l < lanes; ++l) {
+          const uint32_t slot = wave_begin * lanes + l;
+          if (slot >= na) break;
+          never_run.push_back(h_order_[slot]);
+        }
+        ++wave_begin;
+        continue;
+      }
+    }
     uint64_t peq_off = 0, tb_off = 0, s_off = 0;
     uint32_t w = wave_begin;
     uint32_t launch_waves = 0;
     for (; w < num_waves_total; ++w) {
-      uint32_t nb = K, mmax = 0;
-      for (uint32_t l = 0; l < lanes; ++l) {
-        const uint32_t slot = w * lanes + l;
-        if (slot >= na) break;
-        const AlnDesc& d = h_descs_[h_order_[slot]];
-        nb = std::max(nb, (d.q_len + 63) / 64);
-        mmax = std::max(mmax, d.t_len);
-      }
-      const uint64_t peq_need = static_cast<uint64_t>(nb) * 4 * kLanes;
-      const uint64_t tb_need = static_cast<uint64_t>(mmax + 1) * K * 2 * kLanes;
-      const uint64_t s_need = static_cast<uint64_t>(mmax + 1) * K * kLanes;
+      uint64_t peq_need, tb_need, s_need;
+      auto nbm = wave_needs(w, &peq_need, &tb_need, &s_need);
       if (launch_waves > 0 && (peq_off + peq_need > peq_cap_u64_ ||
                                tb_off + tb_need > tb_cap_u64_ || s_off + s_need > s_cap_i32_)) {
         break;
@@ -247,17 +272,14 @@ void AlignerBatch::run() {
       wd.peq_off = peq_off;
       wd.tb_off = tb_off;
       wd.s_off = s_off;
-      wd.nb = nb;
-      wd.mmax = mmax;
+      wd.nb = nbm.first;
+      wd.mmax = nbm.second;
       h_waves_[launch_waves] = wd;
       peq_off += peq_need;
       tb_off += tb_need;
       s_off += s_need;
       ++launch_waves;
     }
-    // a single wave exceeding the arena alone can't be helped; it still has
-    // its own full region (offsets 0) — the caps guarantee this fits because
-    // max_len * K * 2 * 64 * 8 is carved into tb_cap by construction.
     const uint32_t launch_align =
         std::min(na - wave_begin * lanes, launch_waves * lanes);
     RGA_HIP_CHECK(hipMemcpyAsync(d_waves_, h_waves_, launch_waves * sizeof(AlnWaveDesc),
@@ -276,6 +298,11 @@ void AlignerBatch::run() {
   RGA_HIP_CHECK(hipMemcpyAsync(h_status_, arena_.status, na * 4, hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipMemcpyAsync(h_edit_, arena_.edit_distance, na * 4, hipMemcpyDeviceToHost, s));
   RGA_HIP_CHECK(hipStreamSynchronize(s));
+  for (uint32_t orig : never_run) {
+    h_status_[orig] = kAlnNotRun;
+    h_path_len_[orig] = 0;
+    h_edit_[orig] = -1;
+  }
 }
 
 std::string AlignerBatch::cigar_of(uint32_t slot) const {

@@ -105,7 +105,7 @@ class HipPolisher : public Polisher {
       : Polisher(std::move(sparser), std::move(oparser), std::move(tparser), config) {
     int n = hip::device_count();
     if (n < 1) {
-      fprintf(stderr, "[racon::HipPolisher] error: no HIP devices available!\n");
+      fprintf(stderr, "[rga::HipPolisher] error: no HIP devices available!\n");
       exit(1);
     }
     for (int d = 0; d < n; ++d) {
@@ -113,7 +113,7 @@ class HipPolisher : public Polisher {
       RGA_HIP_CHECK(hipFree(nullptr));  // create the context up front
       devices_.emplace_back(d);
     }
-    fprintf(stderr, "[racon::HipPolisher] using %d GPU(s)\n", n);
+    fprintf(stderr, "[rga::HipPolisher] using %d GPU(s)\n", n);
   }
 
   // GPU overlap alignment; overlaps the GPU skips or fails keep an empty
@@ -216,11 +216,11 @@ class HipPolisher : public Polisher {
     batches.clear();
 
     fprintf(stderr,
-            "[racon::HipPolisher] align timings: queue %.3f s, pack+gpu+cigar %.3f s "
+            "[rga::HipPolisher] align timings: queue %.3f s, pack+gpu+cigar %.3f s "
             "(sum over %zu batch threads)\n",
             t_fill_ns.load() / 1e9, t_gpu_ns.load() / 1e9, threads.size());
     if (skipped.load() > 0) {
-      fprintf(stderr, "[racon::HipPolisher] %lu overlap(s) aligned on CPU\n",
+      fprintf(stderr, "[rga::HipPolisher] %lu overlap(s) aligned on CPU\n",
               static_cast<unsigned long>(skipped.load()));
     }
     // CPU pass: walks CIGARs into breaking points; aligns leftovers with the
@@ -240,9 +240,13 @@ class HipPolisher : public Polisher {
           std::max({std::abs(static_cast<int32_t>(config_.match)),
                     std::abs(static_cast<int32_t>(config_.mismatch)),
                     std::abs(static_cast<int32_t>(config_.gap))});
-      if (static_cast<int32_t>(2047 + 1024) * worst_param > 32000) {
+      // thresholds must match the PoaBatch constructor guard exactly: the
+      // kernel clamps stored scores at -28000 in every mode, and banded mode
+      // reserves values below that as out-of-band sentinels
+      if (static_cast<int32_t>(2047 + 1024) * worst_param >
+          (config_.banded_poa ? 27000 : 28000)) {
         fprintf(stderr,
-                "[racon::HipPolisher] warning: score parameters too large for the "
+                "[rga::HipPolisher] warning: score parameters too large for the "
                 "int16 GPU POA scores; polishing windows on the CPU instead\n");
         Polisher::polish(dst, drop_unpolished);
         return;
@@ -255,17 +259,28 @@ class HipPolisher : public Polisher {
     constexpr uint32_t kMaxDepth = 200;  // MAX_DEPTH_PER_WINDOW, cudapolisher.cpp:226
     std::vector<std::unique_ptr<hip::PoaBatch>> batches;
     std::vector<uint64_t> batch_keys;
-    for (int d : devices_) {
-      RGA_HIP_CHECK(hipSetDevice(d));
-      size_t free_mem = 0, total_mem = 0;
-      RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
-      size_t budget = free_mem * 9 / 10 / config_.poa_batches;
-      for (uint32_t b = 0; b < config_.poa_batches; ++b) {
-        batches.emplace_back(acquire_poa(d, budget, config_.match, config_.mismatch,
-                                         config_.gap, config_.banded_poa, kMaxDepth));
-        batch_keys.emplace_back(poa_key(d, config_.match, config_.mismatch, config_.gap,
-                                        config_.banded_poa, kMaxDepth));
+    try {
+      for (int d : devices_) {
+        RGA_HIP_CHECK(hipSetDevice(d));
+        size_t free_mem = 0, total_mem = 0;
+        RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
+        size_t budget = free_mem * 9 / 10 / config_.poa_batches;
+        for (uint32_t b = 0; b < config_.poa_batches; ++b) {
+          batches.emplace_back(acquire_poa(d, budget, config_.match, config_.mismatch,
+                                           config_.gap, config_.banded_poa, kMaxDepth));
+          batch_keys.emplace_back(poa_key(d, config_.match, config_.mismatch, config_.gap,
+                                          config_.banded_poa, kMaxDepth));
+        }
       }
+    } catch (const std::exception& e) {
+      fprintf(stderr,
+              "[rga::HipPolisher] warning: %s; polishing windows on the CPU instead\n",
+              e.what());
+      for (size_t b = 0; b < batches.size(); ++b) {
+        release_poa(batch_keys[b], std::move(batches[b]));
+      }
+      Polisher::polish(dst, drop_unpolished);
+      return;
     }
 
     std::vector<bool> polished(windows_.size(), false);
@@ -334,7 +349,7 @@ class HipPolisher : public Polisher {
     }
     batches.clear();
     fprintf(stderr,
-            "[racon::HipPolisher] poa timings: queue %.3f s, pack+gpu+post %.3f s "
+            "[rga::HipPolisher] poa timings: queue %.3f s, pack+gpu+post %.3f s "
             "(sum over %zu batch threads)\n",
             t_fill_ns.load() / 1e9, t_gpu_ns.load() / 1e9, threads.size());
 
@@ -349,11 +364,11 @@ class HipPolisher : public Polisher {
       }
     }
     if (num_fallback > 0) {
-      fprintf(stderr, "[racon::HipPolisher] %lu window(s) re-polished on CPU\n",
+      fprintf(stderr, "[rga::HipPolisher] %lu window(s) re-polished on CPU\n",
               static_cast<unsigned long>(num_fallback));
       generate_consensus_cpu(polished, &todo);
     } else {
-      logger_->log("[racon::Polisher::polish] generated consensus");
+      logger_->log("[rga::Polisher] generated consensus");
     }
 
     collect(dst, drop_unpolished, polished);

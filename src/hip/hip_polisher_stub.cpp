@@ -17,7 +17,7 @@ namespace hip {
 int runtime_device_count() { return 0; }
 std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
     const std::vector<std::pair<std::string, std::string>>&, uint32_t) {
-  fprintf(stderr, "[racon::hip::align_pairs] error: no HIP backend in this build!\n");
+  fprintf(stderr, "[rga::hip::align_pairs] error: no HIP backend in this build!\n");
   exit(1);
 }
 }  // namespace hip

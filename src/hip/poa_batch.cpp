@@ -5,6 +5,7 @@
 #include <algorithm>
 #include <chrono>
 #include <cstring>
+#include <stdexcept>
 
 #include "hip/hip_common.hpp"
 
@@ -41,12 +42,16 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
                   std::max({std::abs(static_cast<int32_t>(match)),
                             std::abs(static_cast<int32_t>(mismatch)),
                             std::abs(static_cast<int32_t>(gap))});
-  if (worst > (banded ? 27000 : 32000)) {
-    fprintf(stderr,
-            "[rga::hip::PoaBatch] error: score parameters too large for int16 "
-            "GPU scores (|m|,|x|,|g| must keep (%u+%u)*max <= 32000)\n",
-            limits_.max_nodes, limits_.matrix_width);
-    exit(1);
+  // the kernel clamps stored scores at -28000 in every mode, so the usable
+  // magnitude is 28000 (not the int16 32767 limit); banded mode additionally
+  // reserves values below -28000 as out-of-band sentinels
+  if (worst > (banded ? 27000 : 28000)) {
+    char msg[160];
+    snprintf(msg, sizeof(msg),
+             "[rga::hip::PoaBatch] score parameters too large for int16 GPU "
+             "scores (|m|,|x|,|g| must keep (%u+%u)*max <= %d)",
+             limits_.max_nodes, limits_.matrix_width, banded ? 27000 : 28000);
+    throw std::runtime_error(msg);
   }
 
   // The kernel compiles the capacity model as constants (poa_kernel.hip
@@ -60,9 +65,9 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
         limits_.max_ring != defaults.max_ring ||
         limits_.matrix_width != defaults.matrix_width ||
         limits_.max_consensus != defaults.max_consensus) {
-      fprintf(stderr, "[rga::hip::PoaBatch] error: PoaLimits diverged from the "
-                      "compile-time kernel capacity model!\n");
-      exit(1);
+      throw std::runtime_error(
+          "[rga::hip::PoaBatch] PoaLimits diverged from the compile-time "
+          "kernel capacity model");
     }
   }
 
@@ -302,8 +307,13 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     const uint32_t first = h_layer_index_[w];
     return h_layer_ends_[first + h_desc_[w].num_seqs - 1];  // total layer bytes
   };
+  // bucket reads h_desc_, which is overwritten with the sorted descriptors
+  // below — snapshot per original window up front so both the comparator and
+  // the launch-range split see the pre-sort values
+  std::vector<uint32_t> bucket_of(nw0);
+  for (uint32_t i = 0; i < nw0; ++i) bucket_of[i] = bucket(i);
   std::sort(perm.begin(), perm.end(), [&](uint32_t a, uint32_t b) {
-    const uint32_t ba = bucket(a), bb = bucket(b);
+    const uint32_t ba = bucket_of[a], bb = bucket_of[b];
     if (ba != bb) return ba < bb;
     const uint32_t ca = cost(a), cb = cost(b);
     if (ca != cb) return ca > cb;
@@ -333,9 +343,9 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   {
     uint32_t begin = 0;
     while (begin < nw0) {
-      const uint32_t wb = bucket(perm[begin]);
+      const uint32_t wb = bucket_of[perm[begin]];
       uint32_t end = begin + 1;
-      while (end < nw0 && bucket(perm[end]) == wb) {
+      while (end < nw0 && bucket_of[perm[end]] == wb) {
         ++end;
       }
       launch_poa_kernel(arena_, begin, end - begin, wb, stream_);

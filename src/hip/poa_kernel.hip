@@ -957,10 +957,20 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           }
         }
         if (WB == 8 && cbase < len) {
-          // one aligned u64 store covers the lane's 8 move bytes
-          // (cbase is a multiple of 8 exactly when WB == 8); bytes beyond
-          // nown encode columns past the row end and are never read back
-          *reinterpret_cast<uint64_t*>(Mrow + cbase) = mvpack;
+          if (cbase + WB < c.MW) {
+            // one aligned u64 store covers the lane's 8 move bytes
+            // (cbase is a multiple of 8 exactly when WB == 8); bytes beyond
+            // nown encode columns past the row end and are never read back
+            *reinterpret_cast<uint64_t*>(Mrow + cbase) = mvpack;
+          } else {
+            // near-full row: byte MW-1 is the shifted column-0 slot (stored
+            // above as kMvUp); a packed store here would clobber it with the
+            // padding byte 0 (kMvDiag) and derail the traceback at j==0.
+            // Store only the live bytes individually.
+            for (uint32_t w = 0; w < nown; ++w) {
+              Mrow[cbase + w] = static_cast<uint8_t>(mvpack >> (8 * w));
+            }
+          }
         }
         if (len > base && len <= base + kLanes * WB) {
           const int owner = static_cast<int>((len - 1 - base) / WB);

@@ -97,7 +97,8 @@ struct PoaDeviceArena {
 
 // Launches the columns-per-lane kernel variant for windows
 // [window_base, window_base + num_windows) of the (bucket-sorted) desc
-// array. wb must be 5, 9 or 16 and cover ceil(window max_len / 64).
+// array. wb must be 5 or 8 (lane-blocked column widths; the kernel runs
+// multiple passes when wb*64 < the row length).
 void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                        uint32_t num_windows, uint32_t wb, void* stream);
 
