@@ -179,6 +179,149 @@ void append_run(std::string& cigar, uint32_t n, char op) {
   cigar.append(buf, len);
 }
 
+// Scores of the FINAL column of NW(q[0..qn) vs t[0..tn)) for every query
+// prefix length 0..qn: out[i] = NW(q[0..i), t). out[0] = tn (boundary row).
+void nw_last_column(const char* q, uint32_t qn, const char* t, uint32_t tn,
+                    std::vector<int32_t>* out) {
+  out->assign(qn + 1, 0);
+  (*out)[0] = static_cast<int32_t>(tn);
+  if (qn == 0) {
+    return;
+  }
+  Peq peq(q, qn, t, tn);
+  int nb = peq.num_blocks;
+  std::vector<Word> P(nb, ~Word(0));
+  std::vector<Word> M(nb, 0);
+  std::vector<int32_t> bottom(nb);
+  for (int b = 0; b < nb; ++b) {
+    bottom[b] = (b + 1) * kWordBits;
+  }
+  for (uint32_t c = 0; c < tn; ++c) {
+    const Word* eq_col =
+        &peq.eq[static_cast<size_t>(peq.code_of[static_cast<unsigned char>(t[c])]) * nb];
+    int hin = 1;
+    for (int b = 0; b < nb; ++b) {
+      hin = myers_step(P[b], M[b], eq_col[b], hin);
+      bottom[b] += hin;
+    }
+  }
+  // unpack the vertical delta bits into prefix scores (top-down accumulate)
+  int32_t v = static_cast<int32_t>(tn);
+  for (uint32_t i = 0; i < qn; ++i) {
+    const Word mask = Word(1) << (i % kWordBits);
+    const int b = static_cast<int>(i / kWordBits);
+    if (P[b] & mask) {
+      ++v;
+    } else if (M[b] & mask) {
+      --v;
+    }
+    (*out)[i + 1] = v;
+  }
+}
+
+// Backward walk over a fully stored DP (blocked bit-vectors), reproducing
+// the move priority of edlib's obtainAlignmentTraceback: up first ('I',
+// consume query), then left ('D', consume target), then diagonal. Appends
+// ops in FORWARD order.
+void traceback_small(const char* q, uint32_t qn, const char* t, uint32_t tn, std::string* ops) {
+  ColumnStore cs;
+  myers_nw(q, qn, t, tn, &cs);
+
+  std::string rev_ops;
+  rev_ops.reserve(qn + tn);
+  int64_t i = qn - 1;
+  int64_t c = tn - 1;
+  int32_t v = cell_value(cs, c, i);
+  while (i >= 0 || c >= 0) {
+    if (i >= 0) {
+      const int32_t up = cell_value(cs, c, i - 1);
+      if (up + 1 == v) {
+        rev_ops.push_back('I');
+        --i;
+        v = up;
+        continue;
+      }
+    }
+    if (c >= 0) {
+      const int32_t left = cell_value(cs, c - 1, i);
+      if (left + 1 == v) {
+        rev_ops.push_back('D');
+        --c;
+        v = left;
+        continue;
+      }
+    }
+    rev_ops.push_back('M');  // match or mismatch; standard CIGAR merges both
+    v = cell_value(cs, c - 1, i - 1);
+    --i;
+    --c;
+  }
+  ops->append(rev_ops.rbegin(), rev_ops.rend());
+}
+
+// Equal-cost path selection compatible with edlib's obtainAlignment
+// (re-derived from the published algorithm; the reference's edlib submodule
+// is an empty directory — see docs/PARITY.md). Small problems use the
+// stored-DP traceback above; larger ones split the target in half
+// (Hirschberg), score the middle column from both sides, and take the
+// SMALLEST query row whose forward+reverse scores sum to the optimum —
+// that first-crossing rule plus the I>D>M base priority is what pins
+// edlib's choice among equal-cost alignments.
+// rq/rt point at the reverses of q/t (rq[0] == q[qn-1], ...).
+void obtain_ops(const char* q, const char* rq, uint32_t qn, const char* t, const char* rt,
+                uint32_t tn, int64_t score, std::string* ops) {
+  if (qn == 0) {
+    ops->append(tn, 'D');
+    return;
+  }
+  if (tn == 0) {
+    ops->append(qn, 'I');
+    return;
+  }
+
+  // edlib's dispatch rule: full traceback when the stored DP fits in 1 MB
+  // ((2 words + 1 int) per block per target column)
+  static const long long tb_limit = [] {
+    const char* e = getenv("RGA_EDLIB_TB_LIMIT");
+    return e != nullptr ? atoll(e) : 1024ll * 1024;
+  }();
+  const long long num_blocks = (qn + kWordBits - 1) / kWordBits;
+  if ((2ll * sizeof(Word) + sizeof(int32_t)) * num_blocks * tn < tb_limit) {
+    traceback_small(q, qn, t, tn, ops);
+    return;
+  }
+
+  const uint32_t left_w = tn / 2;
+  const uint32_t right_w = tn - left_w;
+
+  // middle-column scores from both directions: fwd[i] = NW(q[0..i), left
+  // half), bwd[j] = NW(q[qn-j..qn), right half)
+  std::vector<int32_t> fwd, bwd;
+  nw_last_column(q, qn, t, left_w, &fwd);
+  nw_last_column(rq, qn, rt, right_w, &bwd);
+
+  // first (lowest) query row where the two halves meet at the optimum
+  int64_t cross = -2;
+  for (uint32_t r = 0; r < qn; ++r) {
+    if (static_cast<int64_t>(fwd[r + 1]) + bwd[qn - 1 - r] == score) {
+      cross = static_cast<int64_t>(r);
+      break;
+    }
+  }
+  if (cross == -2 && static_cast<int64_t>(fwd[0]) + bwd[qn] == score) {
+    cross = -1;  // the optimum consumes the whole left half before any query
+  }
+  if (cross == -2) {
+    // cannot happen for a correct score; fail loudly rather than mis-align
+    fprintf(stderr, "[rga::align] error: no crossing row at the half-target column!\n");
+    exit(1);
+  }
+
+  const uint32_t left_q = static_cast<uint32_t>(cross + 1);
+  obtain_ops(q, rq + (qn - left_q), left_q, t, rt + right_w, left_w, fwd[left_q], ops);
+  obtain_ops(q + left_q, rq, qn - left_q, t + left_w, rt, right_w, bwd[qn - left_q], ops);
+}
+
 }  // namespace
 
 int64_t edit_distance(const char* a, uint32_t a_len, const char* b, uint32_t b_len) {
@@ -202,80 +345,25 @@ std::string align_global_cigar(const char* q, uint32_t q_len, const char* t, uin
     return cigar;
   }
 
-  ColumnStore cs;
-  myers_nw(q, q_len, t, t_len, &cs);
+  const int64_t score = myers_nw(q, q_len, t, t_len, nullptr);
 
-  // Backward walk from (q_len-1, t_len-1) reproducing edlib's move priority.
-  // Priority order is configurable for golden-parity tuning via
-  // RGA_TRACEBACK_ORDER (a permutation of "IDM"); default "IDM" = up ('I',
-  // consume query), then left ('D', consume target), then diagonal ('M').
-  static const char* order_env = getenv("RGA_TRACEBACK_ORDER");
-  const char* order = order_env != nullptr ? order_env : "IDM";
+  std::string rq(q, q + q_len), rt(t, t + t_len);
+  std::reverse(rq.begin(), rq.end());
+  std::reverse(rt.begin(), rt.end());
 
-  std::string ops;  // reversed op chars
+  std::string ops;  // forward 'M'/'I'/'D' chars
   ops.reserve(q_len + t_len);
-  int64_t i = q_len - 1;
-  int64_t c = t_len - 1;
-  int32_t v = cell_value(cs, c, i);
-  while (i >= 0 || c >= 0) {
-    bool moved = false;
-    for (const char* o = order; *o != '\0' && !moved; ++o) {
-      switch (*o) {
-        case 'I':
-          if (i >= 0) {
-            int32_t up = cell_value(cs, c, i - 1);
-            if (up + 1 == v) {
-              ops.push_back('I');
-              --i;
-              v = up;
-              moved = true;
-            }
-          }
-          break;
-        case 'D':
-          if (c >= 0) {
-            int32_t left = cell_value(cs, c - 1, i);
-            if (left + 1 == v) {
-              ops.push_back('D');
-              --c;
-              v = left;
-              moved = true;
-            }
-          }
-          break;
-        case 'M':
-          if (i >= 0 && c >= 0) {
-            int32_t diag = cell_value(cs, c - 1, i - 1);
-            int32_t step = (q[i] == t[c]) ? 0 : 1;
-            if (diag + step == v) {
-              ops.push_back('M');  // match or mismatch; standard CIGAR merges both
-              --i;
-              --c;
-              v = diag;
-              moved = true;
-            }
-          }
-          break;
-      }
-    }
-    if (!moved) {
-      // Fallback diagonal (cannot happen for a consistent DP).
-      ops.push_back('M');
-      --i;
-      --c;
-      v = cell_value(cs, c, i);
-    }
-  }
+  obtain_ops(q, rq.data(), q_len, t, rt.data(), t_len, score, &ops);
 
-  // Collapse the reversed op string into CIGAR runs (forward order).
+  // collapse into CIGAR runs
   uint32_t run = 0;
   char run_op = 0;
-  for (auto it = ops.rbegin(); it != ops.rend(); ++it) {
-    if (*it == run_op) {
+  for (char op : ops) {
+    if (op == run_op) {
       ++run;
     } else {
       append_run(cigar, run, run_op);
-      run_op = *it;
+      run_op = op;
       run = 1;
     }
   }
