@@ -2,16 +2,18 @@
 """Flagship benchmark: Mbp polished per second (whole job), ONT-style
 synthetic long reads, C. elegans-class config (BASELINE.json).
 
-One process per GPU (torchrun / torch.distributed, RCCL over xGMI). Each
-rank polishes a fixed-size shard of the synthetic genome on its own GPU
-(weak scaling: per-GPU work constant; at --gpus 8 the node polishes a full
-C. elegans-sized 100 Mbp genome per step at 30x coverage). A step is the
-whole job: parse, align overlaps, route windows, POA consensus on the GPU,
-and gather of the polished contigs to rank 0.
+One process per GPU. The driver launches N>1 via torch.distributed.run,
+which only supplies RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* env vars — the
+communication itself is the engine's own PyTorch-free component
+(src/hip/comm.cpp): TCP control plane for bootstrap/sizes, RCCL over xGMI
+for the polished-contig gather (length-prefixed point-to-point sends, no
+padded ring). Each rank polishes a fixed-size shard of the synthetic genome
+on its own GPU (weak scaling: per-GPU work constant; at --gpus 8 the node
+polishes a full C. elegans-sized 100 Mbp genome per step at 30x coverage).
+A step is the whole job: parse, align overlaps, route windows, POA
+consensus on the GPU, and gather of the polished contigs to rank 0.
 
 Usage: python bench.py --gpus N --steps K --warmup W
-(N>1 is launched by the driver via torch.distributed.run; RANK/LOCAL_RANK/
-WORLD_SIZE/MASTER_* read from the env.)
 """
 
 import argparse
@@ -31,9 +33,6 @@ sys.path.insert(0, str(REPO))
 # devices and misreport single-GPU throughput.
 LOCAL_RANK = int(os.environ.get("LOCAL_RANK", 0))
 os.environ.setdefault("HIP_VISIBLE_DEVICES", str(LOCAL_RANK))
-
-import torch  # noqa: E402
-import torch.distributed as dist  # noqa: E402
 
 
 def log(msg):
@@ -64,21 +63,26 @@ def main():
     if args.threads is None:
         args.threads = max(8, (os.cpu_count() or 16) // (2 * world))
 
-    have_gpu = torch.cuda.is_available() and not args.cpu
+    import _racon
+
+    have_gpu = _racon.device_count() > 0 and not args.cpu
     # Honest default: the BASELINE config is 12.5 Mbp/GPU (100 Mbp at 8 GPUs).
     # Without a GPU the CPU path cannot finish that in minutes; shrink and say so.
     genome_mbp = args.genome_mbp if args.genome_mbp is not None else (12.5 if have_gpu else 0.3)
     data_tag = "synthetic" if have_gpu and genome_mbp >= 12.5 else "synthetic-reduced"
 
     if world > 1:
-        dist.init_process_group(backend="nccl" if have_gpu else "gloo")
-
-    device = torch.device("cuda:0") if have_gpu else torch.device("cpu")
-    if have_gpu:
-        torch.cuda.set_device(device)
+        # RCCL data plane needs one distinct GPU per rank; a single-GPU
+        # rehearsal (two ranks pinned to the same device) sets
+        # RGA_COMM_FORCE_TCP=1 to exercise the full multi-process flow with
+        # the loopback data plane instead.
+        use_rccl = have_gpu and os.environ.get("RGA_COMM_FORCE_TCP", "0") != "1"
+        host = os.environ.get("MASTER_ADDR", "127.0.0.1")
+        port = int(os.environ.get("RGA_COMM_PORT",
+                                  int(os.environ.get("MASTER_PORT", "29476")) + 41))
+        _racon.comm_init(rank, world, host, port, use_rccl)
 
     from racon_amd import synth
-    import _racon
 
     # -------- setup (untimed): per-rank shard of the synthetic genome --------
     shard_bp = int(genome_mbp * 1e6)
@@ -103,29 +107,22 @@ def main():
                             threads=args.threads, window_length=args.window,
                             poa_batches=poa_batches, aligner_batches=aligner_batches,
                             banded_poa=args.banded)
-        # gather polished contigs to rank 0 (variable-length bytes over RCCL)
+        # gather polished contigs to rank 0: length-prefixed point-to-point
+        # over RCCL/xGMI (comm.cpp) — each rank's exact bytes, no padding
         if world > 1:
             blob = "".join(s for _, s in out).encode()
-            t = torch.frombuffer(bytearray(blob), dtype=torch.uint8).to(device)
-            sizes = torch.zeros(world, dtype=torch.int64, device=device)
-            sizes[rank] = t.numel()
-            dist.all_reduce(sizes)
-            maxlen = int(sizes.max().item())
-            padded = torch.zeros(maxlen, dtype=torch.uint8, device=device)
-            padded[: t.numel()] = t
-            # all_gather rather than gather: gather is not supported on the
-            # NCCL (RCCL) backend; the padded all_gather is one extra hop of
-            # xGMI traffic and works on both backends
-            bufs = [torch.empty(maxlen, dtype=torch.uint8, device=device)
-                    for _ in range(world)]
-            dist.all_gather(bufs, padded)
+            parts = _racon.comm_gather(blob, 0)
+            if rank == 0:
+                assert len(parts) == world
         return sum(len(s) for _, s in out)
 
     def barrier_sync():
+        # barrier + full device sync on both sides of the timed region
+        # (equivalent of dist.barrier() + torch.cuda.synchronize())
         if world > 1:
-            dist.barrier()
+            _racon.comm_barrier()
         if have_gpu:
-            torch.cuda.synchronize()
+            _racon.device_synchronize()
 
     for w in range(args.warmup):
         one_step()
@@ -141,14 +138,8 @@ def main():
 
     # whole-job numbers: MAX time over ranks, SUM of polished bp
     if world > 1:
-        stats = torch.tensor([elapsed, float(polished_bp)], dtype=torch.float64,
-                             device=device if have_gpu else "cpu")
-        tmax = stats[0:1].clone()
-        dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
-        total = stats[1:2].clone()
-        dist.all_reduce(total, op=dist.ReduceOp.SUM)
-        elapsed = float(tmax.item())
-        polished_bp = float(total.item())
+        elapsed = _racon.comm_allreduce_max(elapsed)
+        polished_bp = _racon.comm_allreduce_sum(float(polished_bp))
 
     if rank == 0:
         value = polished_bp / 1e6 / elapsed
@@ -179,7 +170,7 @@ def main():
         }))
 
     if world > 1:
-        dist.destroy_process_group()
+        _racon.comm_finalize()
 
 
 if __name__ == "__main__":

@@ -381,6 +381,8 @@ class HipPolisher : public Polisher {
 namespace hip {
 int runtime_device_count() { return device_count(); }
 
+void runtime_device_synchronize() { RGA_HIP_CHECK(hipDeviceSynchronize()); }
+
 // Direct GPU alignment of raw (query, target) pairs — numerics testing
 // entry (GPU edit distance must equal the CPU optimum; CIGARs must be
 // consistent). Returns (cigar, edit_distance, status) per pair.

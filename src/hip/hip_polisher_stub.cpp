@@ -10,11 +10,34 @@
 #include <vector>
 
 #include "core/polisher.hpp"
+#include "hip/comm.hpp"
 
 namespace rga {
 
+namespace comm {
+Communicator::~Communicator() {}
+void Communicator::init(int, int, const std::string&, int, bool) {
+  fprintf(stderr, "[rga::comm] error: no HIP backend in this build!\n");
+  exit(1);
+}
+void Communicator::finalize() {}
+std::vector<std::string> Communicator::gather(const std::string&, int) { return {}; }
+double Communicator::allreduce_max(double v) { return v; }
+double Communicator::allreduce_sum(double v) { return v; }
+void Communicator::barrier() {}
+void Communicator::ctl_send(int, const void*, size_t) {}
+void Communicator::ctl_recv(int, void*, size_t) {}
+std::vector<std::string> Communicator::ctl_gather(const std::string&, int) { return {}; }
+void Communicator::ctl_bcast(void*, size_t, int) {}
+Communicator& world_comm() {
+  static Communicator c;
+  return c;
+}
+}  // namespace comm
+
 namespace hip {
 int runtime_device_count() { return 0; }
+void runtime_device_synchronize() {}
 std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
     const std::vector<std::pair<std::string, std::string>>&, uint32_t) {
   fprintf(stderr, "[rga::hip::align_pairs] error: no HIP backend in this build!\n");

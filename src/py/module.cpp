@@ -18,9 +18,11 @@
 #include "align/poa.hpp"
 #include "core/polisher.hpp"
 #include "core/sequence.hpp"
+#include "hip/comm.hpp"
 
 namespace rga::hip {
 int runtime_device_count();  // provided by the HIP backend (or the stub)
+void runtime_device_synchronize();
 std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
     const std::vector<std::pair<std::string, std::string>>& pairs, uint32_t band_width);
 }
@@ -123,6 +125,43 @@ PYBIND11_MODULE(_racon, m) {
 
   m.def("device_count", [] { return rga::hip::runtime_device_count(); },
         "Number of visible HIP devices.");
+  m.def("device_synchronize", [] {
+    py::gil_scoped_release release;
+    rga::hip::runtime_device_synchronize();
+  });
+
+  // distributed communicator (hip/comm.hpp): TCP control plane + RCCL data
+  // plane; one process per GPU
+  m.def("comm_init", [](int rank, int world, const std::string& host, int port, bool use_gpu) {
+    py::gil_scoped_release release;
+    rga::comm::world_comm().init(rank, world, host, port, use_gpu);
+  }, py::arg("rank"), py::arg("world"), py::arg("host"), py::arg("port"), py::arg("use_gpu"));
+  m.def("comm_gather", [](const py::bytes& payload, int root) {
+    std::string data = payload;
+    std::vector<std::string> parts;
+    {
+      py::gil_scoped_release release;
+      parts = rga::comm::world_comm().gather(data, root);
+    }
+    py::list out;
+    for (auto& p : parts) {
+      out.append(py::bytes(p));
+    }
+    return out;
+  }, py::arg("payload"), py::arg("root") = 0);
+  m.def("comm_allreduce_max", [](double v) {
+    py::gil_scoped_release release;
+    return rga::comm::world_comm().allreduce_max(v);
+  });
+  m.def("comm_allreduce_sum", [](double v) {
+    py::gil_scoped_release release;
+    return rga::comm::world_comm().allreduce_sum(v);
+  });
+  m.def("comm_barrier", [] {
+    py::gil_scoped_release release;
+    rga::comm::world_comm().barrier();
+  });
+  m.def("comm_finalize", [] { rga::comm::world_comm().finalize(); });
   m.def("edit_distance", &edit_distance_py, py::arg("a"), py::arg("b"));
   m.def("align_cigar", &align_cigar_py, py::arg("query"), py::arg("target"));
   m.def("gpu_align", [](const std::vector<std::pair<std::string, std::string>>& pairs,

@@ -240,3 +240,29 @@ def test_gpu_aligner_sub_launch_split(racon):
     for (q, tt), (cigar, ed, st) in list(zip(pairs, res))[:20]:
         if st == 0:
             assert ed == racon.edit_distance(q, tt)
+
+
+def test_gpu_near_full_width_rows(racon, tmp_path_factory, fasta_reader):
+    """Windows whose DP rows approach the 1024-wide matrix edge (layer
+    length 1017..1023). Regression: the WB=8 packed u64 move store used to
+    overwrite the shifted column-0 move byte at MW-1 for these rows,
+    corrupting the traceback into out-of-bounds device writes."""
+    from racon_amd import synth
+
+    d = tmp_path_factory.mktemp("fullwidth")
+    # insertion-only reads over 1000 bp windows: layer lengths concentrate
+    # at ~1020 +- 5, densely covering the 1017..1023 trigger range
+    s = synth.make_sample(d, genome_bp=30000, coverage=30, seed=33,
+                          sub=0.0, ins=0.02, dele=0.0)
+    truth = list(fasta_reader(s["reference"]).values())[0]
+
+    gpu = racon.polish(s["reads"], s["overlaps"], s["layout"],
+                       threads=4, window_length=1000, poa_batches=1)
+    assert len(gpu) == 1
+    ed = racon.edit_distance(gpu[0][1], truth)
+    assert ed < 0.005 * len(truth), f"corrupted consensus: {ed} errors"
+
+    cpu = racon.polish(s["reads"], s["overlaps"], s["layout"],
+                       threads=4, window_length=1000)
+    ed_cpu_gpu = racon.edit_distance(cpu[0][1], gpu[0][1])
+    assert ed_cpu_gpu < 0.005 * len(cpu[0][1]), ed_cpu_gpu
