@@ -34,9 +34,18 @@ uint32_t pick_band_k(uint32_t band_width) {
 AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
     : device_(device), band_k_(pick_band_k(band_width)) {
   limits_.band = band_k_ * 64;
-  RGA_HIP_CHECK(hipSetDevice(device_));
+  try {
+    allocate_arenas(mem_budget);
+  } catch (...) {
+    release_all();
+    throw;
+  }
+}
+
+void AlignerBatch::allocate_arenas(size_t mem_budget) {
+  RGA_HIP_TRY(hipSetDevice(device_));
   hipStream_t s;
-  RGA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  RGA_HIP_TRY(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   stream_ = s;
 
   // arena split: the per-column band state dominates — per alignment
@@ -44,49 +53,66 @@ AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
   // Cap the device pool so construction stays cheap and the POA arenas
   // (pooled at the same time) keep headroom on 288 GB parts: at K=4 a
   // 20 kbp alignment's state is ~1.3 MB, so 12 GB holds ~9k alignments.
-  const size_t pool = std::min<size_t>(mem_budget, 12ull << 30);
-  seq_cap_ = std::max<size_t>(16u << 20, pool / 96);
-  path_cap_ = seq_cap_;
-  peq_cap_u64_ = seq_cap_ / 8;  // 4 codes per 64 bases = q_bytes/2 of u64s is
-                                // generous; /8 covers the wave-max padding
-  max_alignments_ = 65536;
-
-  const size_t state = pool - 2 * seq_cap_ - peq_cap_u64_ * 8 -
-                       max_alignments_ * (sizeof(AlnDesc) + 16) - (2u << 20);
-  tb_cap_u64_ = state / 20 * 16 / 8;  // 16/20 of state bytes as u64
-  s_cap_i32_ = state / 20 * 4 / 4;    // 4/20 of state bytes as i32
-
-  const uint32_t max_waves = max_alignments_ / kLanes + 2;
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_seqs_), seq_cap_));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_descs_),
-                              max_alignments_ * sizeof(AlnDesc)));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_path_), path_cap_));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_path_len_), max_alignments_ * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_status_), max_alignments_ * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_edit_), max_alignments_ * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_order_), max_alignments_ * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_waves_),
-                              max_waves * sizeof(AlnWaveDesc)));
-
+  // On OOM (shared-device setups) the pool halves and retries.
+  size_t pool = std::min<size_t>(mem_budget, 12ull << 30);
   size_t total = 0;
   auto carve = [&total](size_t bytes) {
     size_t off = total;
     total += (bytes + 255) & ~size_t(255);
     return off;
   };
-  size_t o_seqs = carve(seq_cap_);
-  size_t o_descs = carve(max_alignments_ * sizeof(AlnDesc));
-  size_t o_peq = carve(peq_cap_u64_ * 8);
-  size_t o_tb = carve(tb_cap_u64_ * 8);
-  size_t o_s = carve(s_cap_i32_ * 4);
-  size_t o_path = carve(path_cap_);
-  size_t o_plen = carve(max_alignments_ * 4);
-  size_t o_status = carve(max_alignments_ * 4);
-  size_t o_ed = carve(max_alignments_ * 4);
-  size_t o_order = carve(max_alignments_ * 4);
-  size_t o_waves = carve(max_waves * sizeof(AlnWaveDesc));
+  size_t o_seqs = 0, o_descs = 0, o_peq = 0, o_tb = 0, o_s = 0, o_path = 0, o_plen = 0,
+         o_status = 0, o_ed = 0, o_order = 0, o_waves = 0;
+  uint32_t max_waves = 0;
+  for (;; pool /= 2) {
+    seq_cap_ = std::max<size_t>(16u << 20, pool / 96);
+    path_cap_ = seq_cap_;
+    peq_cap_u64_ = seq_cap_ / 8;  // 4 codes per 64 bases = q_bytes/2 of u64s
+                                  // is generous; /8 covers wave-max padding
+    max_alignments_ = 65536;
 
-  RGA_HIP_CHECK(hipMalloc(&d_pool_, total));
+    const size_t fixed = 2 * seq_cap_ + peq_cap_u64_ * 8 +
+                         max_alignments_ * (sizeof(AlnDesc) + 16) + (2u << 20);
+    const size_t state = pool > fixed ? pool - fixed : (16u << 20);
+    tb_cap_u64_ = state / 20 * 16 / 8;  // 16/20 of state bytes as u64
+    s_cap_i32_ = state / 20 * 4 / 4;    // 4/20 of state bytes as i32
+
+    max_waves = max_alignments_ / kLanes + 2;
+    total = 0;
+    o_seqs = carve(seq_cap_);
+    o_descs = carve(max_alignments_ * sizeof(AlnDesc));
+    o_peq = carve(peq_cap_u64_ * 8);
+    o_tb = carve(tb_cap_u64_ * 8);
+    o_s = carve(s_cap_i32_ * 4);
+    o_path = carve(path_cap_);
+    o_plen = carve(max_alignments_ * 4);
+    o_status = carve(max_alignments_ * 4);
+    o_ed = carve(max_alignments_ * 4);
+    o_order = carve(max_alignments_ * 4);
+    o_waves = carve(max_waves * sizeof(AlnWaveDesc));
+
+    hipError_t err = hipMalloc(&d_pool_, total);
+    if (err == hipSuccess) {
+      break;
+    }
+    d_pool_ = nullptr;
+    (void)hipGetLastError();
+    if (pool <= (256u << 20)) {
+      throw std::runtime_error(
+          "[rga::hip::AlignerBatch] device arena allocation failed (out of memory)");
+    }
+  }
+
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_seqs_), seq_cap_));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_descs_),
+                            max_alignments_ * sizeof(AlnDesc)));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_path_), path_cap_));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_path_len_), max_alignments_ * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_status_), max_alignments_ * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_edit_), max_alignments_ * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_order_), max_alignments_ * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_waves_),
+                            max_waves * sizeof(AlnWaveDesc)));
   auto base = static_cast<uint8_t*>(d_pool_);
   arena_.seqs = base + o_seqs;
   arena_.descs = reinterpret_cast<AlnDesc*>(base + o_descs);
@@ -105,7 +131,9 @@ AlignerBatch::AlignerBatch(int device, size_t mem_budget, uint32_t band_width)
   arena_.limits = limits_;
 }
 
-AlignerBatch::~AlignerBatch() {
+AlignerBatch::~AlignerBatch() { release_all(); }
+
+void AlignerBatch::release_all() {
   (void)hipSetDevice(device_);
   if (d_pool_ != nullptr) {
     (void)hipFree(d_pool_);

@@ -59,6 +59,11 @@ class AlignerBatch {
   void reset();
 
  private:
+  void allocate_arenas(size_t mem_budget);
+  void release_all();
+
+
+ private:
   int device_;
   void* stream_ = nullptr;
   AlnLimits limits_;

@@ -5,6 +5,8 @@
 
 #include <cstdio>
 #include <cstdlib>
+#include <stdexcept>
+#include <string>
 
 namespace rga::hip {
 
@@ -15,6 +17,17 @@ namespace rga::hip {
       fprintf(stderr, "[rga::hip] error: %s failed: %s (%s:%d)\n", #expr,            \
               hipGetErrorString(rga_hip_err_), __FILE__, __LINE__);                  \
       exit(1);                                                                       \
+    }                                                                                \
+  } while (0)
+
+// Throwing variant for recoverable paths (arena construction): callers fall
+// back to smaller arenas or the CPU engine instead of dying.
+#define RGA_HIP_TRY(expr)                                                            \
+  do {                                                                               \
+    hipError_t rga_hip_err_ = (expr);                                                \
+    if (rga_hip_err_ != hipSuccess) {                                                \
+      throw std::runtime_error(std::string("[rga::hip] ") + #expr + " failed: " +    \
+                               hipGetErrorString(rga_hip_err_));                     \
     }                                                                                \
   } while (0)
 

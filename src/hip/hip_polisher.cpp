@@ -96,6 +96,18 @@ void release_poa(uint64_t key, std::unique_ptr<hip::PoaBatch> b) {
   p.poa.emplace_back(key, std::move(b));
 }
 
+// Fraction of free HBM a polish phase may claim (per process). One process
+// per exclusive GPU wants the reference's 0.9; several ranks sharing one
+// device (rehearsals) export RGA_MEM_FRACTION to split it explicitly.
+double mem_fraction() {
+  static const double frac = [] {
+    const char* e = getenv("RGA_MEM_FRACTION");
+    double f = e != nullptr ? atof(e) : 0.9;
+    return f > 0.0 && f <= 0.95 ? f : 0.9;
+  }();
+  return frac;
+}
+
 }  // namespace
 
 class HipPolisher : public Polisher {
@@ -150,15 +162,27 @@ class HipPolisher : public Polisher {
 
     std::vector<std::unique_ptr<hip::AlignerBatch>> batches;
     std::vector<int> batch_devices;
-    for (int d : devices_) {
-      RGA_HIP_CHECK(hipSetDevice(d));
-      size_t free_mem = 0, total_mem = 0;
-      RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
-      size_t budget = free_mem * 9 / 10 / config_.aligner_batches;
-      for (uint32_t b = 0; b < config_.aligner_batches; ++b) {
-        batches.emplace_back(acquire_aligner(d, budget, band));
-        batch_devices.emplace_back(d);
+    try {
+      for (int d : devices_) {
+        RGA_HIP_CHECK(hipSetDevice(d));
+        size_t free_mem = 0, total_mem = 0;
+        RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
+        size_t budget =
+            static_cast<size_t>(free_mem * mem_fraction()) / config_.aligner_batches;
+        for (uint32_t b = 0; b < config_.aligner_batches; ++b) {
+          batches.emplace_back(acquire_aligner(d, budget, band));
+          batch_devices.emplace_back(d);
+        }
       }
+    } catch (const std::exception& e) {
+      fprintf(stderr,
+              "[rga::HipPolisher] warning: %s; aligning overlaps on the CPU instead\n",
+              e.what());
+      for (size_t b = 0; b < batches.size(); ++b) {
+        release_aligner(batch_devices[b], band, std::move(batches[b]));
+      }
+      Polisher::find_overlap_breaking_points(overlaps);
+      return;
     }
 
     std::mutex queue_mutex;
@@ -264,7 +288,7 @@ class HipPolisher : public Polisher {
         RGA_HIP_CHECK(hipSetDevice(d));
         size_t free_mem = 0, total_mem = 0;
         RGA_HIP_CHECK(hipMemGetInfo(&free_mem, &total_mem));
-        size_t budget = free_mem * 9 / 10 / config_.poa_batches;
+        size_t budget = static_cast<size_t>(free_mem * mem_fraction()) / config_.poa_batches;
         for (uint32_t b = 0; b < config_.poa_batches; ++b) {
           batches.emplace_back(acquire_poa(d, budget, config_.match, config_.mismatch,
                                            config_.gap, config_.banded_poa, kMaxDepth));

@@ -71,71 +71,104 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
     }
   }
 
-  RGA_HIP_CHECK(hipSetDevice(device_));
+  RGA_HIP_TRY(hipSetDevice(device_));
   hipStream_t s;
-  RGA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  RGA_HIP_TRY(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   stream_ = s;
 
   const size_t per_window = slab_bytes(limits_) + kSeqArenaPerWindow +
                             limits_.max_consensus * 3 + sizeof(PoaWindowDesc) + 1024;
   num_slabs_ = static_cast<uint32_t>(
       std::min<size_t>(4096, std::max<size_t>(32, mem_budget / per_window)));
-  seq_arena_cap_ = static_cast<size_t>(num_slabs_) * kSeqArenaPerWindow;
 
-  const size_t max_layers = static_cast<size_t>(num_slabs_) * (max_depth_ + 1);
+  // Exception safety: the destructor does not run when the constructor
+  // throws, so free whatever was acquired before rethrowing (the polisher
+  // catches and falls back to the CPU engine).
+  try {
+    allocate_arenas(banded);
+  } catch (...) {
+    release_all();
+    throw;
+  }
+}
 
-  // ---- pinned host staging ----
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_seq_), seq_arena_cap_));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_weight_), seq_arena_cap_));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_layer_ends_), max_layers * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_layer_index_), (num_slabs_ + 1) * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_desc_),
-                              num_slabs_ * sizeof(PoaWindowDesc)));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_),
-                              static_cast<size_t>(num_slabs_) * limits_.max_consensus));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_coverage_),
-                              static_cast<size_t>(num_slabs_) * limits_.max_consensus * 2));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_len_), num_slabs_ * 4));
-  RGA_HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_status_), num_slabs_ * 4));
-
-  // ---- device pool, carved into the arena struct ----
+// ---- device pool + pinned staging. The device pool is the big allocation:
+// on OOM the slab count is halved and the layout retried — shared-device
+// setups (rehearsals, several ranks per GPU) must degrade to smaller
+// batches, not die (the round-1 sizing assumed one exclusive device).
+void PoaBatch::allocate_arenas(bool banded) {
   const PoaLimits& L = limits_;
-  size_t n = L.max_nodes;
+  const size_t n = L.max_nodes;
   size_t total = 0;
   auto carve = [&total](size_t bytes) {
     size_t off = total;
     total += (bytes + 255) & ~size_t(255);
     return off;
   };
-  size_t o_seq = carve(seq_arena_cap_);
-  size_t o_wt = carve(seq_arena_cap_);
-  size_t o_ends = carve(max_layers * 4);
-  size_t o_ends_idx = carve((num_slabs_ + 1) * 4);
-  size_t o_desc = carve(num_slabs_ * sizeof(PoaWindowDesc));
-  size_t o_letters = carve(num_slabs_ * n);
-  size_t o_in_cnt = carve(num_slabs_ * n);
-  size_t o_out_cnt = carve(num_slabs_ * n);
-  size_t o_ring_cnt = carve(num_slabs_ * n);
-  size_t o_in_edges = carve(num_slabs_ * n * L.max_edges * 2);
-  size_t o_in_w = carve(num_slabs_ * n * L.max_edges * 4);
-  size_t o_out_edges = carve(num_slabs_ * n * L.max_edges * 2);
-  size_t o_ring = carve(num_slabs_ * n * L.max_ring * 2);
-  size_t o_nseq = carve(num_slabs_ * n * 2);
-  size_t o_rank = carve(num_slabs_ * n * 2);
-  size_t o_hb_score = carve(num_slabs_ * n * 8);
-  size_t o_hb_pred = carve(num_slabs_ * n * 4);
-  size_t o_aln_n = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
-  size_t o_aln_s = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
-  size_t o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
-  size_t o_moves = carve(num_slabs_ * (n + 1) * L.matrix_width);
-  size_t o_rd = carve(num_slabs_ * n * 8);
-  size_t o_timing = carve(static_cast<size_t>(num_slabs_) * 8 * 8);
-  size_t o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
-  size_t o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
-  size_t o_clen = carve(num_slabs_ * 4);
-  size_t o_status = carve(num_slabs_ * 4);
+  size_t max_layers = 0;
+  size_t o_seq = 0, o_wt = 0, o_ends = 0, o_ends_idx = 0, o_desc = 0, o_letters = 0,
+         o_in_cnt = 0, o_out_cnt = 0, o_ring_cnt = 0, o_in_edges = 0, o_in_w = 0,
+         o_out_edges = 0, o_ring = 0, o_nseq = 0, o_rank = 0, o_hb_score = 0, o_hb_pred = 0,
+         o_aln_n = 0, o_aln_s = 0, o_matrix = 0, o_moves = 0, o_rd = 0, o_timing = 0,
+         o_cons = 0, o_cov = 0, o_clen = 0, o_status = 0;
+  for (;; num_slabs_ /= 2) {
+    seq_arena_cap_ = static_cast<size_t>(num_slabs_) * kSeqArenaPerWindow;
+    max_layers = static_cast<size_t>(num_slabs_) * (max_depth_ + 1);
+    total = 0;
+    o_seq = carve(seq_arena_cap_);
+    o_wt = carve(seq_arena_cap_);
+    o_ends = carve(max_layers * 4);
+    o_ends_idx = carve((num_slabs_ + 1) * 4);
+    o_desc = carve(num_slabs_ * sizeof(PoaWindowDesc));
+    o_letters = carve(num_slabs_ * n);
+    o_in_cnt = carve(num_slabs_ * n);
+    o_out_cnt = carve(num_slabs_ * n);
+    o_ring_cnt = carve(num_slabs_ * n);
+    o_in_edges = carve(num_slabs_ * n * L.max_edges * 2);
+    o_in_w = carve(num_slabs_ * n * L.max_edges * 4);
+    o_out_edges = carve(num_slabs_ * n * L.max_edges * 2);
+    o_ring = carve(num_slabs_ * n * L.max_ring * 2);
+    o_nseq = carve(num_slabs_ * n * 2);
+    o_rank = carve(num_slabs_ * n * 2);
+    o_hb_score = carve(num_slabs_ * n * 8);
+    o_hb_pred = carve(num_slabs_ * n * 4);
+    o_aln_n = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
+    o_aln_s = carve(num_slabs_ * (2 * L.matrix_width + n) * 4);
+    o_matrix = carve(num_slabs_ * (n + 1) * L.matrix_width * 2);
+    o_moves = carve(num_slabs_ * (n + 1) * L.matrix_width);
+    o_rd = carve(num_slabs_ * n * 8);
+    o_timing = carve(static_cast<size_t>(num_slabs_) * 8 * 8);
+    o_cons = carve(static_cast<size_t>(num_slabs_) * L.max_consensus);
+    o_cov = carve(static_cast<size_t>(num_slabs_) * L.max_consensus * 2);
+    o_clen = carve(num_slabs_ * 4);
+    o_status = carve(num_slabs_ * 4);
 
-  RGA_HIP_CHECK(hipMalloc(&d_pool_, total));
+    hipError_t err = hipMalloc(&d_pool_, total);
+    if (err == hipSuccess) {
+      break;
+    }
+    d_pool_ = nullptr;
+    (void)hipGetLastError();  // clear the sticky OOM
+    if (num_slabs_ <= 64) {
+      throw std::runtime_error(
+          "[rga::hip::PoaBatch] device arena allocation failed (out of memory)");
+    }
+  }
+
+  // ---- pinned host staging, sized to the slab count that fit ----
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_seq_), seq_arena_cap_));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_weight_), seq_arena_cap_));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_layer_ends_), max_layers * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_layer_index_), (num_slabs_ + 1) * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_desc_),
+                            num_slabs_ * sizeof(PoaWindowDesc)));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_),
+                            static_cast<size_t>(num_slabs_) * limits_.max_consensus));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_coverage_),
+                            static_cast<size_t>(num_slabs_) * limits_.max_consensus * 2));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_consensus_len_), num_slabs_ * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_status_), num_slabs_ * 4));
+
   auto base = static_cast<uint8_t*>(d_pool_);
   arena_.seq_data = base + o_seq;
   arena_.weight_data = base + o_wt;
@@ -171,7 +204,9 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   arena_.limits = limits_;
 }
 
-PoaBatch::~PoaBatch() {
+PoaBatch::~PoaBatch() { release_all(); }
+
+void PoaBatch::release_all() {
   (void)hipSetDevice(device_);
   if (d_pool_ != nullptr) {
     (void)hipFree(d_pool_);

@@ -45,6 +45,11 @@ class PoaBatch {
   void reset();
 
  private:
+  // Constructor phases (see .cpp): OOM-degrading arena allocation and the
+  // matching cleanup used by both the destructor and a throwing constructor.
+  void allocate_arenas(bool banded);
+  void release_all();
+
   int device_;
   void* stream_ = nullptr;
 
