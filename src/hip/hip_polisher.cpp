@@ -407,6 +407,81 @@ int runtime_device_count() { return device_count(); }
 
 void runtime_device_synchronize() { RGA_HIP_CHECK(hipDeviceSynchronize()); }
 
+// Window-level CPU-vs-GPU differ entry: runs the HIP POA kernel directly on
+// raw windows (backbone first; layer spans in backbone coordinates), so a
+// divergent window found end-to-end can be isolated and replayed. Returns
+// (consensus, polished) per window — exactly what the pipeline would store.
+// Windows the GPU cannot run (< 3 layers, overflow) return the same CPU
+// fallback semantics as HipPolisher::polish.
+std::vector<std::pair<std::string, bool>> poa_windows_gpu(
+    const std::vector<std::vector<std::tuple<std::string, std::string, uint32_t, uint32_t>>>&
+        window_layers,
+    int8_t match, int8_t mismatch, int8_t gap, bool banded, bool trim, bool tgs) {
+  std::vector<std::pair<std::string, bool>> out(window_layers.size());
+  std::vector<std::shared_ptr<Window>> windows;
+  windows.reserve(window_layers.size());
+  for (const auto& layers : window_layers) {
+    if (layers.empty()) {
+      throw std::runtime_error("poa_windows_gpu: window without a backbone");
+    }
+    const auto& bb = layers.front();
+    auto w = createWindow(0, 0, tgs ? WindowType::kTGS : WindowType::kNGS,
+                          std::get<0>(bb).data(),
+                          static_cast<uint32_t>(std::get<0>(bb).size()),
+                          std::get<1>(bb).data(),
+                          static_cast<uint32_t>(std::get<1>(bb).size()));
+    for (size_t i = 1; i < layers.size(); ++i) {
+      const auto& l = layers[i];
+      const std::string& q = std::get<1>(l);
+      w->add_layer(std::get<0>(l).data(), static_cast<uint32_t>(std::get<0>(l).size()),
+                   q.empty() ? nullptr : q.data(), static_cast<uint32_t>(q.size()),
+                   std::get<2>(l), std::get<3>(l));
+    }
+    windows.push_back(std::move(w));
+  }
+
+  hip::PoaBatch batch(0, 4ull << 30, match, mismatch, gap, banded, 200);
+  size_t begin = 0;
+  std::vector<int64_t> slot_of(windows.size(), -1);
+  auto flush = [&](size_t end) {
+    std::vector<bool> status = batch.generate(trim);
+    size_t k = 0;
+    for (size_t i = begin; i < end; ++i) {
+      if (slot_of[i] >= 0) {
+        out[i] = {windows[i]->consensus(), status[k++]};
+      }
+    }
+    batch.reset();
+    begin = end;
+  };
+  size_t next_slot = 0;
+  for (size_t i = 0; i < windows.size(); ++i) {
+    if (windows[i]->num_layers() < 3) {
+      auto bb = windows[i]->sequence(0);
+      out[i] = {std::string(bb.first, bb.second), false};
+      continue;
+    }
+    bool never_fits = false;
+    if (batch.add_window(windows[i], &never_fits)) {
+      slot_of[i] = static_cast<int64_t>(next_slot++);
+      continue;
+    }
+    if (never_fits) {
+      out[i] = {std::string(), false};
+      continue;
+    }
+    flush(i);
+    next_slot = 0;
+    if (batch.add_window(windows[i], &never_fits)) {
+      slot_of[i] = static_cast<int64_t>(next_slot++);
+    } else {
+      out[i] = {std::string(), false};
+    }
+  }
+  flush(windows.size());
+  return out;
+}
+
 // Direct GPU alignment of raw (query, target) pairs — numerics testing
 // entry (GPU edit distance must equal the CPU optimum; CIGARs must be
 // consistent). Returns (cigar, edit_distance, status) per pair.

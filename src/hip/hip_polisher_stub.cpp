@@ -38,6 +38,12 @@ Communicator& world_comm() {
 namespace hip {
 int runtime_device_count() { return 0; }
 void runtime_device_synchronize() {}
+std::vector<std::pair<std::string, bool>> poa_windows_gpu(
+    const std::vector<std::vector<std::tuple<std::string, std::string, uint32_t, uint32_t>>>&,
+    int8_t, int8_t, int8_t, bool, bool, bool) {
+  fprintf(stderr, "[rga::hip::poa_windows_gpu] error: no HIP backend in this build!\n");
+  exit(1);
+}
 std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
     const std::vector<std::pair<std::string, std::string>>&, uint32_t) {
   fprintf(stderr, "[rga::hip::align_pairs] error: no HIP backend in this build!\n");

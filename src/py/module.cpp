@@ -25,6 +25,10 @@ int runtime_device_count();  // provided by the HIP backend (or the stub)
 void runtime_device_synchronize();
 std::vector<std::tuple<std::string, int32_t, int32_t>> align_pairs(
     const std::vector<std::pair<std::string, std::string>>& pairs, uint32_t band_width);
+std::vector<std::pair<std::string, bool>> poa_windows_gpu(
+    const std::vector<std::vector<std::tuple<std::string, std::string, uint32_t, uint32_t>>>&
+        window_layers,
+    int8_t match, int8_t mismatch, int8_t gap, bool banded, bool trim, bool tgs);
 }
 
 namespace py = pybind11;
@@ -109,6 +113,40 @@ std::string poa_consensus(const std::vector<std::string>& seqs,
   return graph.generate_consensus(nullptr);
 }
 
+// CPU side of the window-level differ: the same raw-window inputs as
+// hip::poa_windows_gpu, run through the pipeline's own Window::generate_consensus
+// (subgraph trick, trim, tie-breaks — not the simplified poa_consensus above).
+std::vector<std::pair<std::string, bool>> poa_windows_cpu(
+    const std::vector<std::vector<std::tuple<std::string, std::string, uint32_t, uint32_t>>>&
+        window_layers,
+    int8_t match, int8_t mismatch, int8_t gap, bool trim, bool tgs) {
+  py::gil_scoped_release release;
+  rga::poa::NWEngine engine(match, mismatch, gap);
+  std::vector<std::pair<std::string, bool>> out;
+  out.reserve(window_layers.size());
+  for (const auto& layers : window_layers) {
+    if (layers.empty()) {
+      throw std::runtime_error("poa_windows_cpu: window without a backbone");
+    }
+    const auto& bb = layers.front();
+    auto w = rga::createWindow(0, 0, tgs ? rga::WindowType::kTGS : rga::WindowType::kNGS,
+                               std::get<0>(bb).data(),
+                               static_cast<uint32_t>(std::get<0>(bb).size()),
+                               std::get<1>(bb).data(),
+                               static_cast<uint32_t>(std::get<1>(bb).size()));
+    for (size_t i = 1; i < layers.size(); ++i) {
+      const auto& l = layers[i];
+      const std::string& q = std::get<1>(l);
+      w->add_layer(std::get<0>(l).data(), static_cast<uint32_t>(std::get<0>(l).size()),
+                   q.empty() ? nullptr : q.data(), static_cast<uint32_t>(q.size()),
+                   std::get<2>(l), std::get<3>(l));
+    }
+    bool polished = w->generate_consensus(engine, trim);
+    out.emplace_back(w->consensus(), polished);
+  }
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_racon, m) {
@@ -172,6 +210,20 @@ PYBIND11_MODULE(_racon, m) {
         py::arg("pairs"), py::arg("band_width") = 0,
         "GPU Myers aligner on raw (query, target) pairs -> (cigar, edit_distance, status)");
   m.def("reverse_complement", &reverse_complement, py::arg("sequence"));
+  // window-level CPU-vs-GPU differ: identical raw-window inputs through
+  // both engines (layers = [(seq, qual, begin, end)], slot 0 = backbone)
+  m.def("poa_windows_cpu", &poa_windows_cpu, py::arg("windows"), py::arg("match") = 3,
+        py::arg("mismatch") = -5, py::arg("gap") = -4, py::arg("trim") = true,
+        py::arg("tgs") = true);
+  m.def("poa_windows_gpu", [](const std::vector<std::vector<
+                                  std::tuple<std::string, std::string, uint32_t, uint32_t>>>& w,
+                              int8_t match, int8_t mismatch, int8_t gap, bool banded,
+                              bool trim, bool tgs) {
+    py::gil_scoped_release release;
+    return rga::hip::poa_windows_gpu(w, match, mismatch, gap, banded, trim, tgs);
+  }, py::arg("windows"), py::arg("match") = 3, py::arg("mismatch") = -5,
+     py::arg("gap") = -4, py::arg("banded") = false, py::arg("trim") = true,
+     py::arg("tgs") = true);
   m.def("poa_consensus", &poa_consensus, py::arg("sequences"),
         py::arg("qualities") = std::vector<std::string>(), py::arg("match") = 5,
         py::arg("mismatch") = -4, py::arg("gap") = -8);
