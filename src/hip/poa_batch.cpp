@@ -333,10 +333,13 @@ std::vector<bool> PoaBatch::generate(bool trim) {
     if (arena_.band_width != 0) {
       aw = std::min(aw, arena_.band_width + 64);
     }
-    const uint32_t need = (aw + 63) / 64;
-    // 9-wide single-pass was tried and spills 44-64 B/lane of scratch,
-    // which is catastrophically slow — 8-wide multi-pass covers any width
-    return need <= 5 ? 5u : 8u;
+    // (wb, ring-width) variants — see launch_poa_kernel. 9-wide single-pass
+    // was tried and spills 44-64 B/lane of scratch, which is
+    // catastrophically slow; 8-wide multi-pass covers any width, and the
+    // ring width picks the smallest LDS footprint the rows fit in.
+    if (aw <= 320) return 0;  // WB5, 384-wide ring
+    if (aw <= 575) return 1;  // WB8, 576-wide ring
+    return 2;                 // WB8, full 1024-wide ring
   };
   auto cost = [&](uint32_t w) {
     const uint32_t first = h_layer_index_[w];

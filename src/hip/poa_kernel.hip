@@ -106,30 +106,36 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
          (static_cast<uint64_t>(flags) << 48);
 }
 
-// LDS block state, 9.2 KB = 17 blocks/CU. Cross-batch window co-residency
-// is the dominant throughput lever (a 19 KB variant with full graph-array
-// mirrors made each window ~10% faster but halved residency and lost 20%
-// end to end), so only two things live here: the DP row ring, and — for
-// free, aliased into the same union because the phases never overlap — the
-// Kahn scratch + FIFO queue, which kills the store-to-load round trip the
-// serial sort otherwise does against HBM for every node. The queue doubles
-// as the topological order read by the consensus phase. Everything else
-// stays in the global slabs, hidden by co-residency.
+// LDS block state — cross-batch window co-residency is the dominant
+// throughput lever (a 19 KB variant with full graph-array mirrors made each
+// window ~10% faster but halved residency and lost 20% end to end), so only
+// two things live here: the DP row ring, and — aliased into the same union
+// because the phases never overlap — the Kahn scratch + FIFO queue, which
+// kills the store-to-load round trip the serial sort otherwise does against
+// HBM for every node. The queue doubles as the topological order read by
+// the consensus phase. Everything else stays in the global slabs, hidden by
+// co-residency.
+//
+// The ring width W is a template parameter: the kernel is instantiated per
+// width bucket (384 / 576 / 1024 columns) so the ~500-column windows of a
+// default w=500 run pay ~7 KB of LDS (23 blocks/CU, ~5.7 waves/SIMD)
+// instead of the full-width 10.5 KB (15 blocks/CU, ~3.75/SIMD) — LDS, not
+// VGPRs (60), is the occupancy limiter.
+template <uint32_t W>
 struct Shared {
   union {
-    int16_t ring[kRing][kMaxW];  // DP rows (slot = row % kRing)
+    int16_t ring[kRing][W];  // DP rows (slot = row % kRing)
     struct {
-      uint16_t work[kMaxN];   // Kahn in-degree scratch
+      uint8_t work[kMaxN];    // Kahn in-degree scratch (in-degree < 256)
       uint16_t queue[kMaxN];  // Kahn FIFO == topological order
     } kahn;
   } u;
   // per-layer match bitvectors: bit l of match[c][k] says layer base
   // (1 + k*64 + l) equals code c — turns the per-cell seq comparison into
-  // one register bit test instead of an LDS byte read. One spare word so
+  // one register bit test instead of a byte load. One spare word so
   // cross-word extraction at the last chunk never reads out of bounds.
-  uint64_t match[4][kMaxW / 64 + 1];
+  uint64_t match[4][W / 64 + 1];
   uint64_t rd_block[64];  // row descriptors staged 64 at a time
-  uint8_t seq[kMaxW];
 };
 
 __device__ inline int32_t poa_code(uint8_t b) {
@@ -142,8 +148,7 @@ __device__ inline int32_t poa_code(uint8_t b) {
   }
 }
 
-__device__ inline uint16_t out_edge_of(const struct WindowCtx& c, const Shared& s,
-                                       uint32_t node, uint32_t e);
+__device__ inline uint16_t out_edge_of(const struct WindowCtx& c, uint32_t node, uint32_t e);
 
 struct WindowCtx {
   uint8_t* letters;
@@ -183,15 +188,14 @@ struct WindowCtx {
 
 // ---------- serial (lane 0) graph helpers ----------
 
-__device__ inline uint16_t out_edge_of(const WindowCtx& c, const Shared&, uint32_t node,
-                                       uint32_t e) {
+__device__ inline uint16_t out_edge_of(const WindowCtx& c, uint32_t node, uint32_t e) {
   return c.out_edges[node * c.ME + e];
 }
 
-__device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t b, int32_t w) {
+__device__ inline bool add_edge_d(WindowCtx& c, uint32_t a, uint32_t b, int32_t w) {
   uint32_t n_out = c.out_cnt[a];
   for (uint32_t e = 0; e < n_out; ++e) {
-    if (out_edge_of(c, s, a, e) == b) {
+    if (out_edge_of(c, a, e) == b) {
       uint32_t n_in = c.in_cnt[b];
       for (uint32_t f = 0; f < n_in; ++f) {
         if (c.in_edges[b * c.ME + f] == a) {
@@ -215,7 +219,7 @@ __device__ inline bool add_edge_d(WindowCtx& c, Shared& s, uint32_t a, uint32_t 
   return true;
 }
 
-__device__ inline int32_t add_node_d(WindowCtx& c, Shared& /*s*/, uint8_t letter) {
+__device__ inline int32_t add_node_d(WindowCtx& c, uint8_t letter) {
   if (c.num_nodes >= c.MN || c.num_nodes >= kMaxN) {
     c.status = kPoaNodeOverflow;
     return -1;
@@ -231,7 +235,7 @@ __device__ inline int32_t add_node_d(WindowCtx& c, Shared& /*s*/, uint8_t letter
 
 // Threads the traceback path (stored reversed in aln_*) into the graph.
 // Mirrors Graph::add_alignment (src/align/poa.cpp).
-__device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
+__device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq,
                                 const uint8_t* wts, uint32_t len, int32_t aln_len) {
   // first/last aligned sequence positions
   int32_t first_pos = -1, last_pos = -1;
@@ -249,7 +253,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
   int32_t last_counted = -1;
 
   auto link = [&](int32_t a, int32_t b, int32_t w) {
-    if (!add_edge_d(c, s, a, b, w)) {
+    if (!add_edge_d(c, a, b, w)) {
       return;
     }
     if (a != last_counted) {
@@ -266,7 +270,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
 
   // head chain: seq[0 .. first_pos)
   for (int32_t p = 0; p < first_pos; ++p) {
-    int32_t id = add_node_d(c, s, seq[p]);
+    int32_t id = add_node_d(c, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -285,7 +289,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
     int32_t node = c.aln_nodes[k];
     int32_t new_id;
     if (node == -1) {
-      new_id = add_node_d(c, s, letter);
+      new_id = add_node_d(c, letter);
       if (new_id < 0) return;
     } else if (c.letters[node] == letter) {
       new_id = node;
@@ -300,7 +304,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
         }
       }
       if (new_id == -1) {
-        new_id = add_node_d(c, s, letter);
+        new_id = add_node_d(c, letter);
         if (new_id < 0) return;
         // join the ring: new node linked with node and all its partners
         if (nr >= c.MR) {
@@ -334,7 +338,7 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
 
   // tail chain: seq[last_pos+1 .. len)
   for (int32_t p = (last_pos == -1 ? len : last_pos + 1); p < static_cast<int32_t>(len); ++p) {
-    int32_t id = add_node_d(c, s, seq[p]);
+    int32_t id = add_node_d(c, seq[p]);
     if (id < 0) return;
     if (head != -1) {
       link(head, id, prev_weight + wts[p]);
@@ -349,7 +353,8 @@ __device__ void add_alignment_d(WindowCtx& c, Shared& s, const uint8_t* seq,
 // Kahn topological sort (FIFO, deterministic) over the LDS mirrors. The
 // FIFO queue IS the final topological order (s.u.kahn.queue); rank goes to
 // the global slab (read lane-parallel by build_row_desc).
-__device__ void topo_sort_d(WindowCtx& c, Shared& s) {
+template <class SH>
+__device__ void topo_sort_d(WindowCtx& c, SH& s) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
     s.u.kahn.work[i] = c.in_cnt[i];
@@ -364,7 +369,7 @@ __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
     uint16_t u = s.u.kahn.queue[qhead++];
     uint32_t nout = c.out_cnt[u];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t v = out_edge_of(c, s, u, e);
+      uint16_t v = out_edge_of(c, u, e);
       if (--s.u.kahn.work[v] == 0) {
         s.u.kahn.queue[qtail++] = v;
       }
@@ -377,7 +382,8 @@ __device__ void topo_sort_d(WindowCtx& c, Shared& s) {
 
 // Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
 // Returns consensus length written into out/cov (forward order), or -1.
-__device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* cov,
+template <class SH>
+__device__ int32_t consensus_d(WindowCtx& c, SH& s, uint8_t* out, uint16_t* cov,
                                uint32_t max_out) {
   uint32_t n = c.num_nodes;
   for (uint32_t i = 0; i < n; ++i) {
@@ -422,7 +428,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
     // invalidate alternative branches
     uint32_t nout = c.out_cnt[max_id];
     for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t endn = out_edge_of(c, s, max_id, e);
+      uint16_t endn = out_edge_of(c, max_id, e);
       uint32_t nin = c.in_cnt[endn];
       for (uint32_t f = 0; f < nin; ++f) {
         uint16_t o = c.in_edges[endn * c.ME + f];
@@ -490,7 +496,7 @@ __device__ int32_t consensus_d(WindowCtx& c, Shared& s, uint8_t* out, uint16_t* 
 
 // Lane-parallel: rebuild the packed per-rank row descriptors from the graph
 // arrays (called after backbone init and after every topo sort).
-__device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
+__device__ void build_row_desc(WindowCtx& c, int lane) {
   const uint32_t n = c.num_nodes;
   for (uint32_t node = lane; node < n; node += kLanes) {
     const uint32_t r = c.rank[node];
@@ -502,7 +508,7 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
     const uint32_t nout = c.out_cnt[node];
     uint8_t flags = (nout == 0) ? kRdEnd : 0;
     for (uint32_t e = 0; e < nout; ++e) {
-      const uint32_t sr = c.rank[out_edge_of(c, s, node, e)];
+      const uint32_t sr = c.rank[out_edge_of(c, node, e)];
       if (sr - r >= kRing) {
         flags |= kRdStore;
         break;
@@ -514,10 +520,11 @@ __device__ void build_row_desc(WindowCtx& c, Shared& s, int lane) {
 
 // ---------- the mega-kernel ----------
 
-template <bool TIMED, uint32_t WB>
+template <bool TIMED, uint32_t WB, uint32_t MAXW>
 __launch_bounds__(kLanes, 4)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                                   uint32_t num_windows) {
+  static_assert(MAXW <= kMaxW, "ring width exceeds the slab matrix width");
   if (blockIdx.x >= num_windows) {
     return;
   }
@@ -531,7 +538,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
   constexpr PoaLimits L{};
   const uint32_t slab = desc.scratch_idx;
 
-  __shared__ Shared s;
+  __shared__ Shared<MAXW> s;
 
   WindowCtx c;
   c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
@@ -623,16 +630,20 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     const uint32_t len = c.ends[layer] - beg;
     const uint8_t* seq = c.seq_base + beg;
     const uint8_t* wts = c.weight_base + beg;
-    if (len == 0 || len + 1 > c.MW) {
+    if (len == 0) {
       continue;  // host should have filtered; skip defensively
     }
-
-    for (uint32_t j = lane; j < len; j += kLanes) {
-      s.seq[j] = seq[j];
+    if (len + 1 > c.MW || len + 1 > MAXW) {
+      if (len + 1 > c.MW) {
+        continue;  // over the slab matrix: host drops these (defensive)
+      }
+      c.status = kPoaWidthOverflow;  // mis-bucketed: fail to the CPU path
+      break;
     }
-    __syncthreads();
-    // build the per-layer match bitvectors from the LDS copy (lane-parallel
-    // over chunk words; one spare zero word for cross-word extraction)
+
+    // build the per-layer match bitvectors straight from the (L2-resident)
+    // global layer bytes, lane-parallel over chunk words; one spare zero
+    // word so cross-word extraction never reads out of bounds
     {
       const uint32_t words = (len + kLanes - 1) / kLanes + 1;
       for (uint32_t w = lane; w < words * 4; w += kLanes) {
@@ -642,7 +653,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         if (base < len) {
           const uint32_t lim = min(kLanes, len - base);
           for (uint32_t l = 0; l < lim; ++l) {
-            if (poa_code(s.seq[base + l]) == static_cast<int32_t>(cc)) {
+            if (poa_code(seq[base + l]) == static_cast<int32_t>(cc)) {
               bits |= 1ull << l;
             }
           }
@@ -786,7 +797,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           }
         } else if (nown > 0) {
           for (uint32_t w = 0; w < nown; ++w) {
-            if (s.seq[cbase + w] == letter) {
+            if (seq[cbase + w] == letter) {  // rare: non-ACGT row letter
               mbits |= 1ull << w;
             }
           }
@@ -847,9 +858,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 okmask |= 1u;  // col 0 sits outside [plo+1, phi] by construction
               }
               const uint32_t colmax = c.MW - 1;
+              constexpr uint32_t kRingMax = MAXW - 1;  // LDS ring row bound
               if (r + 1 - p < kRing) {
                 const uint32_t slot = p % kRing;
-                if (cbase + WB <= colmax) {
+                if (cbase + WB <= kRingMax) {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
                     const int32_t val = s.u.ring[slot][cbase + w];
@@ -858,7 +870,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 } else {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
-                    const int32_t val = s.u.ring[slot][min(cbase + w, colmax)];
+                    const int32_t val = s.u.ring[slot][min(cbase + w, kRingMax)];
                     pv[w] = ((okmask >> w) & 1u) ? val : kNegInf;
                   }
                 }
@@ -1048,7 +1060,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
 
       t_tb += lap();
       if (c.status == kPoaOk) {
-        add_alignment_d(c, s, seq, wts, len, aln_len);
+        add_alignment_d(c, seq, wts, len, aln_len);
       }
       t_add += lap();
       if (c.status == kPoaOk) {
@@ -1067,7 +1079,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     // rebuild the packed row descriptors for the grown graph, lane-parallel
     (void)lap();
     if (c.status == kPoaOk) {
-      build_row_desc(c, s, lane);
+      build_row_desc(c, lane);
     }
     __syncthreads();
     t_rd += lap();
@@ -1101,36 +1113,44 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
 }  // namespace
 
 void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
-                       uint32_t num_windows, uint32_t wb, void* stream) {
+                       uint32_t num_windows, uint32_t bucket, void* stream) {
   static const bool timed = getenv("RGA_POA_TIMING") != nullptr;
   auto st = static_cast<hipStream_t>(stream);
   const dim3 grid(num_windows), block(kLanes);
-  // separate __global__ instantiations per columns-per-lane bucket: one
-  // kernel containing all variants pays the widest variant's registers on
-  // every path (measured 3x occupancy collapse), separate kernels do not
+  // separate __global__ instantiations per (columns-per-lane, ring-width)
+  // bucket: one kernel containing all variants pays the widest variant's
+  // registers and LDS on every path (measured 3x occupancy collapse).
   // Measured: the power-of-two 8-wide variant (two passes for a 530-column
   // window) beats a 9-wide single-pass variant by ~8% — non-power-of-two
   // unrolls lose more in generated address code than the extra, mostly
-  // empty pass costs. 5-wide serves the banded (-b) window.
+  // empty pass costs. The ring width (LDS) sets occupancy; see Shared<W>.
   if (timed) {
-    switch (wb) {
-      case 5:
-        hipLaunchKernelGGL((poa_window_kernel<true, 5>), grid, block, 0, st, arena,
+    switch (bucket) {
+      case 0:
+        hipLaunchKernelGGL((poa_window_kernel<true, 5, 384>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 1:
+        hipLaunchKernelGGL((poa_window_kernel<true, 8, 576>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
       default:
-        hipLaunchKernelGGL((poa_window_kernel<true, 8>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<true, 8, 1024>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
     }
   } else {
-    switch (wb) {
-      case 5:
-        hipLaunchKernelGGL((poa_window_kernel<false, 5>), grid, block, 0, st, arena,
+    switch (bucket) {
+      case 0:
+        hipLaunchKernelGGL((poa_window_kernel<false, 5, 384>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 1:
+        hipLaunchKernelGGL((poa_window_kernel<false, 8, 576>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
       default:
-        hipLaunchKernelGGL((poa_window_kernel<false, 8>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<false, 8, 1024>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
     }

@@ -32,6 +32,7 @@ enum PoaStatus : int32_t {
   kPoaRingOverflow = 3,
   kPoaConsensusOverflow = 4,
   kPoaNotRun = 5,
+  kPoaWidthOverflow = 6,  // layer wider than the launched ring variant
 };
 
 // Per-window input descriptor (layers already sorted: backbone first, then
@@ -95,11 +96,11 @@ struct PoaDeviceArena {
   PoaLimits limits;
 };
 
-// Launches the columns-per-lane kernel variant for windows
-// [window_base, window_base + num_windows) of the (bucket-sorted) desc
-// array. wb must be 5 or 8 (lane-blocked column widths; the kernel runs
-// multiple passes when wb*64 < the row length).
+// Launches one (columns-per-lane, LDS-ring-width) kernel variant for
+// windows [window_base, window_base + num_windows) of the (bucket-sorted)
+// desc array. bucket: 0 = WB5/384-wide (rows <= 320 columns, single pass),
+// 1 = WB8/576-wide (<= 575), 2 = WB8/1024-wide (the rest, multi-pass).
 void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
-                       uint32_t num_windows, uint32_t wb, void* stream);
+                       uint32_t num_windows, uint32_t bucket, void* stream);
 
 }  // namespace rga::hip
