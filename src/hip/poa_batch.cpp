@@ -329,17 +329,30 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   std::vector<uint32_t> perm(nw0);
   for (uint32_t i = 0; i < nw0; ++i) perm[i] = i;
   auto bucket = [&](uint32_t w) -> uint32_t {
-    uint32_t aw = h_desc_[w].max_len;
+    // (wb, ring-width) variants — see launch_poa_kernel. The LDS ring is
+    // indexed by ABSOLUTE column, so its width follows the true longest
+    // row (aw); only the columns-per-pass choice (WB) follows the banded
+    // clamp. 9-wide single-pass was tried and spills 44-64 B/lane of
+    // scratch, which is catastrophically slow; 8-wide multi-pass covers
+    // any width, and the ring width picks the smallest LDS footprint the
+    // rows fit in.
+    const uint32_t aw = h_desc_[w].max_len;
+    uint32_t pass_w = aw;
     if (arena_.band_width != 0) {
-      aw = std::min(aw, arena_.band_width + 64);
+      pass_w = std::min(pass_w, arena_.band_width + 64);
     }
-    // (wb, ring-width) variants — see launch_poa_kernel. 9-wide single-pass
-    // was tried and spills 44-64 B/lane of scratch, which is
-    // catastrophically slow; 8-wide multi-pass covers any width, and the
-    // ring width picks the smallest LDS footprint the rows fit in.
+    const bool narrow_pass = pass_w <= 320;
     if (aw <= 320) return 0;  // WB5, 384-wide ring
-    if (aw <= 575) return 1;  // WB8, 576-wide ring
-    return 2;                 // WB8, full 1024-wide ring
+    if (aw <= 575) {
+      if (narrow_pass) return 3;  // banded: WB5 passes, 576-wide ring
+      return h_desc_[w].num_seqs <= 96
+                 ? 1   // WB8, 576-wide ring, 1536-node Kahn (deep windows
+                       // can outgrow the smaller node cap; route them to
+                       // the full variant instead of bouncing via the CPU)
+                 : 2;  // WB8, full-width ring, full Kahn
+    }
+    return narrow_pass ? 4   // banded: WB5 passes, full 1024-wide ring
+                       : 2;  // WB8, full 1024-wide ring
   };
   auto cost = [&](uint32_t w) {
     const uint32_t first = h_layer_index_[w];

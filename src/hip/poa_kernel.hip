@@ -121,13 +121,13 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
 // default w=500 run pay ~7 KB of LDS (23 blocks/CU, ~5.7 waves/SIMD)
 // instead of the full-width 10.5 KB (15 blocks/CU, ~3.75/SIMD) — LDS, not
 // VGPRs (60), is the occupancy limiter.
-template <uint32_t W>
+template <uint32_t W, uint32_t MN>
 struct Shared {
   union {
     int16_t ring[kRing][W];  // DP rows (slot = row % kRing)
     struct {
-      uint8_t work[kMaxN];    // Kahn in-degree scratch (in-degree < 256)
-      uint16_t queue[kMaxN];  // Kahn FIFO == topological order
+      uint8_t work[MN];    // Kahn in-degree scratch (in-degree < 256)
+      uint16_t queue[MN];  // Kahn FIFO == topological order
     } kahn;
   } u;
   // per-layer match bitvectors: bit l of match[c][k] says layer base
@@ -520,11 +520,12 @@ __device__ void build_row_desc(WindowCtx& c, int lane) {
 
 // ---------- the mega-kernel ----------
 
-template <bool TIMED, uint32_t WB, uint32_t MAXW>
+template <bool TIMED, uint32_t WB, uint32_t MAXW, uint32_t MAXN = kMaxN>
 __launch_bounds__(kLanes, 4)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                                   uint32_t num_windows) {
   static_assert(MAXW <= kMaxW, "ring width exceeds the slab matrix width");
+  static_assert(MAXN <= kMaxN, "node cap exceeds the slab graph capacity");
   if (blockIdx.x >= num_windows) {
     return;
   }
@@ -538,7 +539,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
   constexpr PoaLimits L{};
   const uint32_t slab = desc.scratch_idx;
 
-  __shared__ Shared<MAXW> s;
+  __shared__ Shared<MAXW, MAXN> s;
 
   WindowCtx c;
   c.letters = a.letters + static_cast<size_t>(slab) * L.max_nodes;
@@ -581,7 +582,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
   c.ME = L.max_edges;
   c.MR = L.max_ring;
   c.MW = L.matrix_width;
-  c.MN = L.max_nodes;
+  c.MN = min(L.max_nodes, MAXN - 1);  // variant Kahn capacity (see Shared)
   c.bw = a.band_width;
   c.m = a.match;
   c.x = a.mismatch;
@@ -1131,7 +1132,15 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                            window_base, num_windows);
         break;
       case 1:
-        hipLaunchKernelGGL((poa_window_kernel<true, 8, 576>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<true, 8, 576, 1536>), grid, block, 0, st,
+                           arena, window_base, num_windows);
+        break;
+      case 3:
+        hipLaunchKernelGGL((poa_window_kernel<true, 5, 576>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 4:
+        hipLaunchKernelGGL((poa_window_kernel<true, 5, 1024>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
       default:
@@ -1146,7 +1155,15 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                            window_base, num_windows);
         break;
       case 1:
-        hipLaunchKernelGGL((poa_window_kernel<false, 8, 576>), grid, block, 0, st, arena,
+        hipLaunchKernelGGL((poa_window_kernel<false, 8, 576, 1536>), grid, block, 0, st,
+                           arena, window_base, num_windows);
+        break;
+      case 3:
+        hipLaunchKernelGGL((poa_window_kernel<false, 5, 576>), grid, block, 0, st, arena,
+                           window_base, num_windows);
+        break;
+      case 4:
+        hipLaunchKernelGGL((poa_window_kernel<false, 5, 1024>), grid, block, 0, st, arena,
                            window_base, num_windows);
         break;
       default:

@@ -99,7 +99,9 @@ struct PoaDeviceArena {
 // Launches one (columns-per-lane, LDS-ring-width) kernel variant for
 // windows [window_base, window_base + num_windows) of the (bucket-sorted)
 // desc array. bucket: 0 = WB5/384-wide (rows <= 320 columns, single pass),
-// 1 = WB8/576-wide (<= 575), 2 = WB8/1024-wide (the rest, multi-pass).
+// 1 = WB8/576-wide/1536-node (<= 575 and depth <= 96), 2 = WB8/1024-wide,
+// 3 = WB5/576-wide (banded rows in a 576 matrix), 4 = WB5/1024-wide
+// (banded rows in a full-width matrix).
 void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                        uint32_t num_windows, uint32_t bucket, void* stream);
 
