@@ -798,6 +798,18 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         }
       }
 
+      // Cross-pass lane-0 sources for the register fast path, captured
+      // BEFORE this row's own saves overwrite prev_hi: pass k's lane 0
+      // needs the predecessor row's column (k*64*WB), which is lane 63's
+      // last value of pass k-1.
+      int32_t cross0[kPasses];
+      if (fast_row) {
+        cross0[0] = prev_h0;
+#pragma unroll
+        for (uint32_t k = 1; k < kPasses; ++k) {
+          cross0[k] = __builtin_amdgcn_readlane(prev_hi[k - 1], kLanes - 1);
+        }
+      }
       if (!banded) {
         prev_h0 = h0;  // consumed by the next row's register fast path
       }
@@ -865,10 +877,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             pv[1 + 2 * q] = static_cast<int32_t>(static_cast<int16_t>(pk & 0xffffu));
             pv[2 + 2 * q] = static_cast<int32_t>(static_cast<int16_t>(pk >> 16));
           }
-          const int32_t lane0_val =
-              (pass == 0) ? prev_h0
-                          : __builtin_amdgcn_readlane(prev_hi[pass - 1], kLanes - 1);
-          pv[0] = __builtin_amdgcn_update_dpp(lane0_val, prev_hi[pass], 0x138, 0xf, 0xf,
+          pv[0] = __builtin_amdgcn_update_dpp(cross0[pass], prev_hi[pass], 0x138, 0xf, 0xf,
                                               false);  // wave_shr:1 = lane l-1
 #pragma unroll
           for (uint32_t w = 0; w < WB; ++w) {
@@ -982,7 +991,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         int32_t hsave[WB];
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
-          hsave[w] = -28000;  // padding: the clamp floor (see fast path)
+          hsave[w] = -32768;  // padding: int16 min, below every valid cell
+                              // (valid cells clamp at -28000)
         }
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
