@@ -675,24 +675,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     uint32_t best_row = 0;
     (void)lap();
 
-    // Register-carried previous row (WB==8, non-banded): most graph rows
-    // are chains whose single predecessor is the immediately preceding
-    // rank, so the predecessor values are exactly what this lane computed
-    // last iteration — serve them from registers (one DPP shift for the
-    // cross-lane column) instead of the LDS ring, and skip the per-row
-    // lgkmcnt drain. This removes the LDS round trip from the serial
-    // row-to-row critical path (the kernel is ~66% wait-parked; see
-    // profiles/). Padding/overflow columns hold the -28000 clamp floor,
-    // which is exactly what the ring stores for clamped cells.
-    constexpr uint32_t kPasses = (kLanes * WB >= MAXW) ? 1u
-        : (MAXW - 1 + kLanes * WB - 1) / (kLanes * WB);
-    uint32_t prev_pk[kPasses][WB / 2 == 0 ? 1 : WB / 2];  // packed int16 pairs
-    int32_t prev_hi[kPasses];  // own last column (col cbase+WB) per pass
-    int32_t prev_h0 = 0;       // column 0 of the previous row (wave-uniform)
-    (void)prev_pk;
-    (void)prev_hi;
-    (void)prev_h0;
-
     // row 0 (all-gap) is arithmetic: H0[j] = j * g — never materialized.
     // Row descriptors are staged 64 at a time through LDS with one
     // coalesced load: a per-row dependent global read (~600 ns) was the
@@ -720,12 +702,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
 
       // predecessor rows (e < kMaxPre precomputed; beyond that re-derived
       // in the chunk loop — nodes with >8 in-edges are rare)
-      const bool fast_row = (WB == 8) && !banded && nin == 1 && pred0 == r && r > 0;
-      // rows that read the LDS ring must see every pending ring store;
-      // register-served rows (and nin==0 rows) skip the drain entirely
-      if (!fast_row && nin != 0) {
-        wave_lds_sync();
-      }
       uint32_t pred_rows[kMaxPre];
       pred_rows[0] = pred0;
       const uint32_t npre = min(nin, kMaxPre);
@@ -772,8 +748,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         int32_t best0 = kNegInf;
         if (nin == 0) {
           best0 = 0;
-        } else if (fast_row) {
-          best0 = prev_h0;  // single pred = the row just computed
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
@@ -796,22 +770,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           // so each lane's 8 move bytes form one aligned u64 store)
           Mrow[c.MW - 1] = static_cast<uint8_t>(kMvUp | (e0 << 2));
         }
-      }
-
-      // Cross-pass lane-0 sources for the register fast path, captured
-      // BEFORE this row's own saves overwrite prev_hi: pass k's lane 0
-      // needs the predecessor row's column (k*64*WB), which is lane 63's
-      // last value of pass k-1.
-      int32_t cross0[kPasses];
-      if (fast_row) {
-        cross0[0] = prev_h0;
-#pragma unroll
-        for (uint32_t k = 1; k < kPasses; ++k) {
-          cross0[k] = __builtin_amdgcn_readlane(prev_hi[k - 1], kLanes - 1);
-        }
-      }
-      if (!banded) {
-        prev_h0 = h0;  // consumed by the next row's register fast path
       }
 
       // ---- lane-blocked columns ----
@@ -862,28 +820,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               bd[w] = (j - 1) * c.g + sub;
               bu[w] = j * c.g + c.g;
             }
-          }
-        } else if (fast_row) {
-          // predecessor = the row this lane computed last iteration: own
-          // columns come straight from registers, the one cross-lane column
-          // from a DPP shift of the neighbor's last value. Padding columns
-          // hold the -28000 clamp floor — never above any valid candidate,
-          // and their results are discarded by the nown guard below.
-          const uint32_t pass = base / (kLanes * WB);
-          int32_t pv[WB + 1];
-#pragma unroll
-          for (uint32_t q = 0; q < WB / 2; ++q) {
-            const uint32_t pk = prev_pk[pass][q];
-            pv[1 + 2 * q] = static_cast<int32_t>(static_cast<int16_t>(pk & 0xffffu));
-            pv[2 + 2 * q] = static_cast<int32_t>(static_cast<int16_t>(pk >> 16));
-          }
-          pv[0] = __builtin_amdgcn_update_dpp(cross0[pass], prev_hi[pass], 0x138, 0xf, 0xf,
-                                              false);  // wave_shr:1 = lane l-1
-#pragma unroll
-          for (uint32_t w = 0; w < WB; ++w) {
-            const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
-            bd[w] = pv[w] + sub;
-            bu[w] = pv[w + 1] + c.g;
           }
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
@@ -988,12 +924,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         // finalize own columns: h, moves, stores
         int32_t h_sel = kNegInf;
         uint64_t mvpack = 0;  // 8 move bytes -> one aligned store at Mrow[cbase]
-        int32_t hsave[WB];
-#pragma unroll
-        for (uint32_t w = 0; w < WB; ++w) {
-          hsave[w] = -32768;  // padding: int16 min, below every valid cell
-                              // (valid cells clamp at -28000)
-        }
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
           if (w < nown) {
@@ -1025,7 +955,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               mv = static_cast<uint8_t>(type | (esel << 2));
             }
             const int32_t h16 = h < -28000 ? -28000 : h;
-            hsave[w] = h16;
             if (store_row) {
               Hrow[j] = static_cast<int16_t>(h16);
             }
@@ -1056,16 +985,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             }
           }
         }
-        if (WB == 8 && !banded) {
-          // carry this row for the next row's register fast path
-          const uint32_t pass = base / (kLanes * WB);
-#pragma unroll
-          for (uint32_t q = 0; q < WB / 2; ++q) {
-            prev_pk[pass][q] = (static_cast<uint32_t>(hsave[2 * q]) & 0xffffu) |
-                               (static_cast<uint32_t>(hsave[2 * q + 1]) << 16);
-          }
-          prev_hi[pass] = hsave[WB - 1];
-        }
         if (len > base && len <= base + kLanes * WB) {
           const int owner = static_cast<int>((len - 1 - base) / WB);
           const int32_t lc = __builtin_amdgcn_readlane(h_sel, owner);
@@ -1073,8 +992,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         }
       }
 
-      // (the per-row lgkmcnt drain moved to the top of the row body and
-      // only runs for rows that actually read the LDS ring)
+      // ring writes must be visible to every lane before the next row;
+      // deliberately NOT __syncthreads (would drain the global row stores)
+      wave_lds_sync();
 
       // end-node max (strict >, first in topological order wins);
       // last_col_val is already wave-uniform (readlane)
