@@ -520,8 +520,13 @@ __device__ void build_row_desc(WindowCtx& c, int lane) {
 
 // ---------- the mega-kernel ----------
 
-template <bool TIMED, uint32_t WB, uint32_t MAXW, uint32_t MAXN = kMaxN>
-__launch_bounds__(kLanes, 4)
+// MINWAVES is the occupancy the compiler must budget registers for
+// (waves/SIMD floor): the 8-wide variants otherwise help themselves to the
+// full 128-VGPR budget of 4 waves/SIMD and registers — not LDS — become
+// the residency limiter.
+template <bool TIMED, uint32_t WB, uint32_t MAXW, uint32_t MAXN = kMaxN,
+          uint32_t MINWAVES = 4>
+__launch_bounds__(kLanes, MINWAVES)
 __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                                   uint32_t num_windows) {
   static_assert(MAXW <= kMaxW, "ring width exceeds the slab matrix width");
@@ -1132,7 +1137,7 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                            window_base, num_windows);
         break;
       case 1:
-        hipLaunchKernelGGL((poa_window_kernel<true, 8, 576, 1536>), grid, block, 0, st,
+        hipLaunchKernelGGL((poa_window_kernel<true, 8, 576, 1536, 5>), grid, block, 0, st,
                            arena, window_base, num_windows);
         break;
       case 3:
@@ -1155,7 +1160,7 @@ void launch_poa_kernel(const PoaDeviceArena& arena, uint32_t window_base,
                            window_base, num_windows);
         break;
       case 1:
-        hipLaunchKernelGGL((poa_window_kernel<false, 8, 576, 1536>), grid, block, 0, st,
+        hipLaunchKernelGGL((poa_window_kernel<false, 8, 576, 1536, 5>), grid, block, 0, st,
                            arena, window_base, num_windows);
         break;
       case 3:
