@@ -305,3 +305,29 @@ def test_gpu_window_differ_matches_cpu(racon):
         bad = [i for i, (c, g) in enumerate(zip(cpu, gpu)) if c[0] != g[0]]
         assert bad == [], [(i, len(cpu[i][0]), len(gpu[i][0]),
                             racon.edit_distance(cpu[i][0], gpu[i][0])) for i in bad]
+
+
+def test_gpu_real_scale_quality_and_determinism(racon, tmp_path_factory, fasta_reader):
+    """Real-scale gate (reference analog: ci/gpu/cuda_test.sh byte-diffs a
+    2.72 Mbp ONT contig): a 3 Mbp synthetic contig polished on the GPU must
+    (a) land within an error bound of the truth and (b) produce byte-exact
+    identical FASTA across thread counts and batch layouts."""
+    from racon_amd import synth
+
+    d = tmp_path_factory.mktemp("realscale")
+    s = synth.make_sample(d, genome_bp=3_000_000, coverage=30, seed=77)
+    truth = list(fasta_reader(s["reference"]).values())[0]
+
+    base = racon.polish(s["reads"], s["overlaps"], s["layout"],
+                        threads=8, poa_batches=2, aligner_batches=2)
+    assert len(base) == 1
+    ed = racon.edit_distance(base[0][1], truth)
+    # draft has ~2% errors; polished must be far below 0.1%
+    assert ed < 0.001 * len(truth), f"quality gate: {ed} errors on 3 Mbp"
+
+    # determinism across configurations: window consensus depends only on
+    # window content, never on batch/thread layout
+    for kw in (dict(threads=4, poa_batches=1, aligner_batches=1),
+               dict(threads=16, poa_batches=4, aligner_batches=4)):
+        other = racon.polish(s["reads"], s["overlaps"], s["layout"], **kw)
+        assert other == base, f"non-deterministic output under {kw}"
