@@ -149,14 +149,20 @@ class HipPolisher : public Polisher {
         }
       }
       if (count > 0) {
-        // 10% of mean span like the reference, clamped to 256 (K=4): the
-        // Myers band is exact inside the band, indel drift on long reads is
-        // a ~sqrt random walk (sd ~30 bp at 20 kbp / 6% error, so ±128 is
-        // >4 sigma), the CPU fallback catches escapes per item — and the
-        // aligner launch is only ~1.5 waves/CU, so per-wave work is wall
-        // time one-for-one
-        band = static_cast<uint32_t>(total_len / count / 10) & ~1u;
+        // 10% of mean span like the reference, clamped scale-aware: the
+        // Myers band is exact inside the band and escapes fall back per
+        // item to the exact CPU aligner, so the clamp is a throughput
+        // knob, not a quality one (measured, docs/DESIGN.md "Band clamp"):
+        // at <=20 kbp spans / 6% error a 256 band completes 24/24 overlaps
+        // exactly; at 100 kbp it escapes 5/24 (6% err) and 13/24 (12%),
+        // all of which the 512 band completes exactly. Wide explicit bands
+        // stay available via --cudaaligner-band-width.
+        const uint64_t mean_span = total_len / count;
+        band = static_cast<uint32_t>(mean_span / 10) & ~1u;
         band = band < 64 ? 64 : (band > 256 ? 256 : band);
+        if (mean_span >= 40000) {
+          band = 512;  // long-read drift outruns the 256 band (see above)
+        }
       }
     }
 
