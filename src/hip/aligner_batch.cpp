@@ -372,7 +372,7 @@ uint32_t AlignerBatch::align_and_emit() {
   // (the run-length walk over ~30 kbp paths x thousands of alignments is
   // otherwise a serial tail after every batch round)
   const size_t n = overlaps_.size();
-  const uint32_t nthreads = std::min<uint32_t>(4, std::max<uint32_t>(1, n / 512));
+  const uint32_t nthreads = std::min<uint32_t>(8, std::max<uint32_t>(1, n / 512));
   std::atomic<uint32_t> failed{0};
   auto emit_range = [&](size_t begin, size_t end) {
     uint32_t local_failed = 0;
