@@ -122,7 +122,7 @@ __device__ inline uint64_t pack_rd(uint8_t letter, uint8_t nin, uint16_t node,
 // instead of the full-width 10.5 KB (15 blocks/CU, ~3.75/SIMD) — LDS, not
 // VGPRs (60), is the occupancy limiter.
 template <uint32_t W, uint32_t MN>
-struct Shared {
+struct alignas(16) Shared {
   union {
     int16_t ring[kRing][W];  // DP rows (slot = row % kRing)
     struct {
@@ -786,6 +786,8 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
       int32_t carry_u = h0;  // u-space max through column j0 (h0 = -inf when
                              // the band excludes column 0)
       int32_t last_col_val = kNegInf;
+      int32_t pass_tail = 0;  // lane 63's last clamped value of the previous
+                              // pass (the shifted store's column `base`)
 
       for (uint32_t base = j0; base < jend; base += kLanes * WB) {
         const uint32_t cbase = base + lane * WB;  // own cols: cbase+1..cbase+WB
@@ -865,9 +867,29 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               }
               const uint32_t colmax = c.MW - 1;
               constexpr uint32_t kRingMax = MAXW - 1;  // LDS ring row bound
+              // unpack a 16-byte vector load into pv[0..7] (cbase*2 is a
+              // multiple of 16 when WB == 8, and both the ring rows and the
+              // global matrix rows are 16-byte aligned); replaces 8 scalar
+              // b16 loads — the LDS pipe is the congested unit at 20+
+              // co-resident windows per CU (profiles/PARKED.md)
+              auto unpack8 = [&pv](int4 v) {
+                const int32_t ws[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+                for (uint32_t q = 0; q < 4; ++q) {
+                  pv[2 * q] = static_cast<int32_t>(static_cast<int16_t>(ws[q] & 0xffff));
+                  pv[2 * q + 1] = ws[q] >> 16;  // arithmetic: sign-extended
+                }
+              };
               if (r + 1 - p < kRing) {
                 const uint32_t slot = p % kRing;
-                if (cbase + WB <= kRingMax) {
+                if (WB == 8 && cbase + WB <= kRingMax) {
+                  unpack8(*reinterpret_cast<const int4*>(&s.u.ring[slot][cbase]));
+                  pv[WB] = s.u.ring[slot][cbase + WB];
+#pragma unroll
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    pv[w] = ((okmask >> w) & 1u) ? pv[w] : kNegInf;
+                  }
+                } else if (cbase + WB <= kRingMax) {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
                     const int32_t val = s.u.ring[slot][cbase + w];
@@ -882,7 +904,14 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
                 }
               } else {
                 const int16_t* gsrc = c.matrix + static_cast<size_t>(p) * c.MW;
-                if (cbase + WB <= colmax) {
+                if (WB == 8 && cbase + WB <= colmax) {
+                  unpack8(*reinterpret_cast<const int4*>(gsrc + cbase));
+                  pv[WB] = gsrc[cbase + WB];
+#pragma unroll
+                  for (uint32_t w = 0; w <= WB; ++w) {
+                    pv[w] = ((okmask >> w) & 1u) ? pv[w] : kNegInf;
+                  }
+                } else if (cbase + WB <= colmax) {
 #pragma unroll
                   for (uint32_t w = 0; w <= WB; ++w) {
                     const int32_t val = gsrc[cbase + w];
@@ -926,15 +955,19 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         // next pass's carry: wave-uniform max over everything <= this pass
         carry_u = max(carry_u, __builtin_amdgcn_readlane(incl, kLanes - 1));
 
-        // finalize own columns: h, moves, stores
+        // finalize own columns: h, moves, stores. h is computed for every
+        // slot (junk past the row end feeds only junk columns); the move
+        // logic stays guarded.
         int32_t h_sel = kNegInf;
         uint64_t mvpack = 0;  // 8 move bytes -> one aligned store at Mrow[cbase]
+        int32_t harr[WB];
 #pragma unroll
         for (uint32_t w = 0; w < WB; ++w) {
+          const uint32_t j = cbase + 1 + w;
+          const int32_t v = max(bd[w], bu[w]);
+          const int32_t h = max(us[w], excl) + static_cast<int32_t>(j) * c.g;
+          harr[w] = h < -28000 ? -28000 : h;
           if (w < nown) {
-            const uint32_t j = cbase + 1 + w;
-            const int32_t v = max(bd[w], bu[w]);
-            const int32_t h = max(us[w], excl) + static_cast<int32_t>(j) * c.g;
             uint8_t mv;
             if (h != v) {
               mv = kMvLeft;
@@ -959,18 +992,47 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               }
               mv = static_cast<uint8_t>(type | (esel << 2));
             }
-            const int32_t h16 = h < -28000 ? -28000 : h;
-            if (store_row) {
-              Hrow[j] = static_cast<int16_t>(h16);
-            }
-            ring_row[j] = static_cast<int16_t>(h16);
             if (WB == 8) {
               mvpack |= static_cast<uint64_t>(mv) << (8 * w);
             } else {
+              if (store_row) {
+                Hrow[j] = static_cast<int16_t>(harr[w]);
+              }
+              ring_row[j] = static_cast<int16_t>(harr[w]);
               Mrow[j - 1] = mv;  // shifted layout, per-byte for odd widths
             }
             if (j == len) {
               h_sel = h;
+            }
+          }
+        }
+        if (WB == 8) {
+          // Shifted 16-byte row stores: lane l writes columns
+          // [cbase .. cbase+7] = [neighbor's last value | own h[0..6]] so
+          // the store is one aligned b128 instead of 8 b16 ops (LDS-pipe
+          // congestion is the bottleneck — profiles/PARKED.md). Column
+          // `base` comes from lane 63 of the previous pass (or h0), the
+          // pass-boundary column base+512 from the NEXT pass's lane 0, and
+          // a row ending exactly on the boundary stores it explicitly.
+          const int32_t tail_in = (base == j0) ? h0 : pass_tail;
+          const int32_t shifted =
+              __builtin_amdgcn_update_dpp(tail_in, harr[WB - 1], 0x138, 0xf, 0xf, false);
+          pass_tail = __builtin_amdgcn_readlane(harr[WB - 1], kLanes - 1);
+          if (cbase < jend) {
+            int4 vec;
+            vec.x = (shifted & 0xffff) | (harr[0] << 16);
+            vec.y = (harr[1] & 0xffff) | (harr[2] << 16);
+            vec.z = (harr[3] & 0xffff) | (harr[4] << 16);
+            vec.w = (harr[5] & 0xffff) | (harr[6] << 16);
+            *reinterpret_cast<int4*>(&ring_row[cbase]) = vec;
+            if (store_row) {
+              *reinterpret_cast<int4*>(Hrow + cbase) = vec;
+            }
+          }
+          if (lane == 0 && len == base + kLanes * WB) {
+            ring_row[len] = static_cast<int16_t>(pass_tail);
+            if (store_row) {
+              Hrow[len] = static_cast<int16_t>(pass_tail);
             }
           }
         }
