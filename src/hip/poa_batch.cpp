@@ -197,6 +197,10 @@ void PoaBatch::allocate_arenas(bool banded) {
   arena_.coverage = reinterpret_cast<uint16_t*>(base + o_cov);
   arena_.consensus_len = reinterpret_cast<uint32_t*>(base + o_clen);
   arena_.status = reinterpret_cast<int32_t*>(base + o_status);
+  arena_.vstore_mode = [] {
+    const char* e = getenv("RGA_VSTORE");
+    return e != nullptr ? static_cast<uint32_t>(atoi(e)) : 1u;
+  }();
   arena_.match = match_;
   arena_.mismatch = mismatch_;
   arena_.gap = gap_;

@@ -994,6 +994,12 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             }
             if (WB == 8) {
               mvpack |= static_cast<uint64_t>(mv) << (8 * w);
+              if (a.vstore_mode != 1) {
+                if (store_row) {
+                  Hrow[j] = static_cast<int16_t>(harr[w]);
+                }
+                ring_row[j] = static_cast<int16_t>(harr[w]);
+              }
             } else {
               if (store_row) {
                 Hrow[j] = static_cast<int16_t>(harr[w]);
@@ -1006,7 +1012,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             }
           }
         }
-        if (WB == 8) {
+        if (WB == 8 && a.vstore_mode != 0) {
           // Shifted 16-byte row stores: lane l writes columns
           // [cbase .. cbase+7] = [neighbor's last value | own h[0..6]] so
           // the store is one aligned b128 instead of 8 b16 ops (LDS-pipe
@@ -1029,10 +1035,22 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               *reinterpret_cast<int4*>(Hrow + cbase) = vec;
             }
           }
-          if (lane == 0 && len == base + kLanes * WB) {
-            ring_row[len] = static_cast<int16_t>(pass_tail);
-            if (store_row) {
-              Hrow[len] = static_cast<int16_t>(pass_tail);
+          // When this pass's last covered column is `len` itself AND len
+          // lands on a lane boundary ((len - base) % WB == 0), the lane
+          // that would store it via its shifted slot has cbase == len and
+          // is excluded by the cbase < jend guard — store it explicitly
+          // from the owning lane's last value. (Missing this for ANY
+          // multiple-of-8 row length left ring[len] uninitialized and made
+          // results depend on stale LDS — caught by tools/probe_order.py.)
+          const uint32_t rem = len > base ? len - base : 0;
+          if (rem > 0 && rem <= kLanes * WB && (rem & (WB - 1)) == 0) {
+            const int owner = static_cast<int>(rem / WB) - 1;
+            const int32_t t16 = __builtin_amdgcn_readlane(harr[WB - 1], owner);
+            if (lane == 0) {
+              ring_row[len] = static_cast<int16_t>(t16);
+              if (store_row) {
+                Hrow[len] = static_cast<int16_t>(t16);
+              }
             }
           }
         }

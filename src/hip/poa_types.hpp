@@ -91,6 +91,7 @@ struct PoaDeviceArena {
   int32_t* status;          // [1] per window
 
   int8_t match, mismatch, gap;
+  uint32_t vstore_mode;  // debug: 0 per-element row stores, 1 vector, 2 both
   uint32_t band_width;  // 0 = full-width DP; else static band (reference -b:
                         // BatchConfig band 256, src/cuda/cudabatch.cpp:56-59)
   PoaLimits limits;
