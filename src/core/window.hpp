@@ -59,6 +59,10 @@ class Window {
   std::pair<const char*, uint32_t> quality(uint32_t i) const {
     return {layers_[i].qual, layers_[i].qual_len};
   }
+  // Backbone span of a routed layer (begin, inclusive end); (0, 0) = backbone.
+  std::pair<uint32_t, uint32_t> span(uint32_t i) const {
+    return {layers_[i].begin, layers_[i].end};
+  }
 
   // Indices of all layers with the backbone first and the rest ordered by
   // start position. Shared by the CPU and HIP consensus paths so a window's

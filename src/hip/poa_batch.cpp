@@ -106,6 +106,7 @@ void PoaBatch::allocate_arenas(bool banded) {
     return off;
   };
   size_t max_layers = 0;
+  size_t o_spans = 0;
   size_t o_seq = 0, o_wt = 0, o_ends = 0, o_ends_idx = 0, o_desc = 0, o_letters = 0,
          o_in_cnt = 0, o_out_cnt = 0, o_ring_cnt = 0, o_in_edges = 0, o_in_w = 0,
          o_out_edges = 0, o_ring = 0, o_nseq = 0, o_rank = 0, o_hb_score = 0, o_hb_pred = 0,
@@ -118,6 +119,7 @@ void PoaBatch::allocate_arenas(bool banded) {
     o_seq = carve(seq_arena_cap_);
     o_wt = carve(seq_arena_cap_);
     o_ends = carve(max_layers * 4);
+    o_spans = carve(max_layers * 4);
     o_ends_idx = carve((num_slabs_ + 1) * 4);
     o_desc = carve(num_slabs_ * sizeof(PoaWindowDesc));
     o_letters = carve(num_slabs_ * n);
@@ -159,6 +161,7 @@ void PoaBatch::allocate_arenas(bool banded) {
   RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_seq_), seq_arena_cap_));
   RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_weight_), seq_arena_cap_));
   RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_layer_ends_), max_layers * 4));
+  RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_layer_span_), max_layers * 4));
   RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_layer_index_), (num_slabs_ + 1) * 4));
   RGA_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&h_desc_),
                             num_slabs_ * sizeof(PoaWindowDesc)));
@@ -173,6 +176,7 @@ void PoaBatch::allocate_arenas(bool banded) {
   arena_.seq_data = base + o_seq;
   arena_.weight_data = base + o_wt;
   arena_.layer_ends = reinterpret_cast<uint32_t*>(base + o_ends);
+  arena_.layer_spans = reinterpret_cast<uint32_t*>(base + o_spans);
   arena_.layer_ends_index = reinterpret_cast<uint32_t*>(base + o_ends_idx);
   arena_.windows = reinterpret_cast<PoaWindowDesc*>(base + o_desc);
   arena_.letters = base + o_letters;
@@ -216,7 +220,8 @@ void PoaBatch::release_all() {
     (void)hipFree(d_pool_);
   }
   for (void* p : {static_cast<void*>(h_seq_), static_cast<void*>(h_weight_),
-                  static_cast<void*>(h_layer_ends_), static_cast<void*>(h_layer_index_),
+                  static_cast<void*>(h_layer_ends_), static_cast<void*>(h_layer_span_),
+                  static_cast<void*>(h_layer_index_),
                   static_cast<void*>(h_desc_), static_cast<void*>(h_consensus_),
                   static_cast<void*>(h_coverage_), static_cast<void*>(h_consensus_len_),
                   static_cast<void*>(h_status_)}) {
@@ -292,6 +297,10 @@ void PoaBatch::pack() {
     uint32_t rel_end = 0;
     uint32_t packed = 0;
     const uint32_t total_layers = window->num_layers();
+    const uint32_t bb_len = window->sequence(0).second;
+    // CPU spanning rule (Window::generate_consensus): a layer within 1% of
+    // both window edges aligns against the full graph
+    const uint32_t edge_margin = static_cast<uint32_t>(0.01 * bb_len);
     for (uint32_t k = 0; k < total_layers && packed < desc.num_seqs; ++k) {
       uint32_t i = order[k];
       auto seq = window->sequence(i);
@@ -299,6 +308,16 @@ void PoaBatch::pack() {
       if (k > 0 && seq.second + 1 > limits_.matrix_width) {
         continue;
       }
+      uint32_t span_word = 0xFFFFFFFFu;  // backbone / spanning layers: full DP
+      if (k > 0) {
+        auto sp = window->span(i);
+        const bool spans_window =
+            sp.first < edge_margin && sp.second > bb_len - edge_margin;
+        if (!spans_window) {
+          span_word = (sp.first << 16) | (sp.second & 0xffffu);
+        }
+      }
+      h_layer_span_[ends_at] = span_word;
       std::memcpy(h_seq_ + off, seq.first, seq.second);
       if (qual.first != nullptr) {
         for (uint32_t b = 0; b < seq.second; ++b) {
@@ -391,6 +410,7 @@ std::vector<bool> PoaBatch::generate(bool trim) {
   d(arena_.seq_data, h_seq_, seq_bytes_);
   d(arena_.weight_data, h_weight_, seq_bytes_);
   d(arena_.layer_ends, h_layer_ends_, num_layer_ends_ * 4);
+  d(arena_.layer_spans, h_layer_span_, num_layer_ends_ * 4);
   d(arena_.layer_ends_index, h_layer_index_, windows_.size() * 4);
   d(arena_.windows, h_desc_, windows_.size() * sizeof(PoaWindowDesc));
 

@@ -63,6 +63,7 @@ class PoaBatch {
   uint8_t* h_seq_ = nullptr;
   uint8_t* h_weight_ = nullptr;
   uint32_t* h_layer_ends_ = nullptr;
+  uint32_t* h_layer_span_ = nullptr;
   uint32_t* h_layer_index_ = nullptr;
   PoaWindowDesc* h_desc_ = nullptr;
   uint8_t* h_consensus_ = nullptr;

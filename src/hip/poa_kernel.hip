@@ -171,7 +171,9 @@ struct WindowCtx {
 
   const uint8_t* seq_base;
   const uint8_t* weight_base;
-  const uint32_t* ends;  // layer end offsets (relative to window start)
+  const uint32_t* ends;   // layer end offsets (relative to window start)
+  const uint32_t* spans;  // per layer: begin<<16|end backbone span, or
+                          // 0xFFFFFFFF for window-spanning layers (full DP)
   uint32_t num_seqs;
 
   uint32_t ME;  // max edges
@@ -583,6 +585,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
   c.seq_base = a.seq_data + desc.seq_offset;
   c.weight_base = a.weight_data + desc.seq_offset;
   c.ends = a.layer_ends + a.layer_ends_index[win];
+  c.spans = a.layer_spans + a.layer_ends_index[win];
   c.num_seqs = desc.num_seqs;
   c.ME = L.max_edges;
   c.MR = L.max_ring;
@@ -671,10 +674,31 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
 
     const uint32_t n = c.num_nodes;
     const uint32_t chunks = (len + kLanes - 1) / kLanes;
-    // banded mode (-b): static band around the rank diagonal
+
+    // SUBGRAPH restriction (CPU parity, Window::generate_consensus): a
+    // layer that does not span the window aligns only against the rank
+    // window [rank(backbone[begin]), rank(backbone[end])]. Rows outside are
+    // skipped, predecessor edges from below rank(begin) are ignored (rows
+    // losing every predecessor become NW start rows), and the alignment
+    // target is the span's end rank (plus true sinks inside the window).
+    // The reference's cudapoa aligns every layer to the full graph instead,
+    // which is what degrades its long-window GPU goldens (racon_test.cpp
+    // pins 4168 vs CPU 1289 at w=1000); this engine keeps the CPU
+    // windowing semantics on device.
+    const uint32_t span_word = c.spans[layer];
+    const bool sub = span_word != 0xFFFFFFFFu;
+    uint32_t rlo = 0, rhi = n - 1;
+    if (sub) {
+      rlo = c.rank[span_word >> 16];
+      rhi = c.rank[span_word & 0xffffu];
+    }
+
+    // banded mode (-b): static band around the rank diagonal of the
+    // (possibly rank-restricted) row window
     const bool banded = (c.bw != 0) && (c.bw < len);
     const uint32_t slope16 =
-        banded ? static_cast<uint32_t>((static_cast<uint64_t>(len) << 16) / n) : 0;
+        banded ? static_cast<uint32_t>((static_cast<uint64_t>(len) << 16) / (rhi - rlo + 1))
+               : 0;
 
     int32_t best_score = kNegInf;
     uint32_t best_row = 0;
@@ -685,13 +709,13 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     // coalesced load: a per-row dependent global read (~600 ns) was the
     // dominant per-row latency and neither scan nor store restructuring
     // moved it (measured via RGA_POA_TIMING).
-    for (uint32_t rblk = 0; rblk < n; rblk += kLanes) {
+    for (uint32_t rblk = rlo & ~(kLanes - 1); rblk <= rhi; rblk += kLanes) {
       if (rblk + lane < n) {
         s.rd_block[lane] = c.row_desc[rblk + lane];
       }
       wave_lds_sync();
-      const uint32_t rlim = min(n, rblk + kLanes);
-    for (uint32_t r = rblk; r < rlim; ++r) {
+      const uint32_t rlim = min(rhi + 1, rblk + kLanes);
+    for (uint32_t r = max(rblk, rlo); r < rlim; ++r) {
       const uint64_t rd = s.rd_block[r - rblk];
       const uint8_t letter = static_cast<uint8_t>(rd);
       const int32_t letter_code = poa_code(letter);
@@ -714,9 +738,27 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         pred_rows[e] = c.rank[c.in_edges[node * c.ME + e]] + 1;
       }
 
+      // usable predecessor edges: in subgraph mode edges from ranks below
+      // the window are cut; a row losing every predecessor becomes an NW
+      // start row (virtual row 0), matching the CPU subgraph's sources.
+      // Edge index 63 in a move byte encodes that virtual start for the
+      // traceback (real indices stop at max_edges-1 = 47).
+      uint64_t usable = (1ull << nin) - 1;  // nin <= 48 < 64
+      if (sub && nin != 0) {
+        usable = 0;
+        for (uint32_t e = 0; e < nin; ++e) {
+          const uint32_t p = (e < kMaxPre) ? pred_rows[e]
+                                           : c.rank[c.in_edges[node * c.ME + e]] + 1;
+          if (p > rlo) {
+            usable |= 1ull << e;
+          }
+        }
+      }
+      const bool vstart = usable == 0;  // nin == 0, or every pred cut
+
       uint32_t row_klo = 0, row_khi = chunks - 1;
       if (banded) {
-        band_chunks(r + 1, slope16, c.bw, chunks, &row_klo, &row_khi);
+        band_chunks(r + 1 - rlo, slope16, c.bw, chunks, &row_klo, &row_khi);
       }
 
       // fetch a predecessor row value: LDS ring if close, global otherwise;
@@ -727,14 +769,14 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         }
         if (banded && col != 0) {
           uint32_t pklo, pkhi;
-          band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+          band_chunks(p - rlo, slope16, c.bw, chunks, &pklo, &pkhi);
           const uint32_t kc = (col - 1) / 64;
           if (kc < pklo || kc > pkhi) {
             return kNegInf;
           }
         } else if (banded && col == 0) {
           uint32_t pklo, pkhi;
-          band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+          band_chunks(p - rlo, slope16, c.bw, chunks, &pklo, &pkhi);
           if (pklo != 0) {
             return kNegInf;
           }
@@ -751,10 +793,14 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
       uint32_t e0 = 0;
       if (row_klo == 0) {
         int32_t best0 = kNegInf;
-        if (nin == 0) {
+        if (vstart) {
           best0 = 0;
+          e0 = 63;  // virtual-start sentinel (see `usable`)
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
+            if (!((usable >> e) & 1)) {
+              continue;
+            }
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
                                              : c.rank[c.in_edges[node * c.ME + e]] + 1;
             const int32_t hp0 = pred_val(p, 0);
@@ -818,18 +864,21 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
           bu[w] = kNegInf;
         }
 
-        if (nin == 0) {
+        if (vstart) {
 #pragma unroll
           for (uint32_t w = 0; w < WB; ++w) {
             if (w < nown) {
               const int32_t j = static_cast<int32_t>(cbase + 1 + w);
-              const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
-              bd[w] = (j - 1) * c.g + sub;
+              const int32_t subst = ((mbits >> w) & 1) ? c.m : c.x;
+              bd[w] = (j - 1) * c.g + subst;
               bu[w] = j * c.g + c.g;
             }
           }
         } else {
           for (uint32_t e = 0; e < nin; ++e) {
+            if (!((usable >> e) & 1)) {
+              continue;
+            }
             const uint32_t p = (e < kMaxPre) ? pred_rows[e]
                                              : c.rank[c.in_edges[node * c.ME + e]] + 1;
             // gather pred row values pv[w] = H(p, cbase + w), w in 0..WB.
@@ -971,7 +1020,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             uint8_t mv;
             if (h != v) {
               mv = kMvLeft;
-            } else if (nin == 0 || nin == 1) {
+            } else if (vstart) {
+              mv = static_cast<uint8_t>(((bd[w] >= bu[w]) ? kMvDiag : kMvUp) | (63u << 2));
+            } else if (nin == 1) {
               mv = (bd[w] >= bu[w]) ? kMvDiag : kMvUp;
             } else {
               // rare multi-pred row: recover the argmax edge (first e wins)
@@ -980,6 +1031,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               uint32_t esel = 0;
               const int32_t sub = ((mbits >> w) & 1) ? c.m : c.x;
               for (uint32_t e = 0; e < nin; ++e) {
+                if (!((usable >> e) & 1)) {
+                  continue;
+                }
                 const uint32_t p = (e < kMaxPre)
                                        ? pred_rows[e]
                                        : c.rank[c.in_edges[node * c.ME + e]] + 1;
@@ -1082,8 +1136,10 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
       wave_lds_sync();
 
       // end-node max (strict >, first in topological order wins);
-      // last_col_val is already wave-uniform (readlane)
-      if (is_end && last_col_val > best_score) {
+      // last_col_val is already wave-uniform (readlane). In subgraph mode
+      // the span's end rank is the alignment sink (plus true sinks inside
+      // the window).
+      if ((is_end || (sub && r == rhi)) && last_col_val > best_score) {
         best_score = last_col_val;
         best_row = r + 1;
       }
@@ -1118,8 +1174,9 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
             rec_seq = static_cast<int32_t>(j - 1);
             prev_j = j - 1;
           } else {
-            uint32_t p = 0;
-            if (nin != 0) {
+            uint32_t p = 0;  // virtual row 0: genuine start rows (nin==0)
+                             // and subgraph sources (edge sentinel 63)
+            if (nin != 0 && e != 63) {
               p = (e == 0) ? static_cast<uint32_t>((rd >> 32) & 0xffff)
                            : c.rank[c.in_edges[node * c.ME + e]] + 1;
             }

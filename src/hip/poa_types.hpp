@@ -51,6 +51,14 @@ struct PoaDeviceArena {
   const uint8_t* seq_data;     // concatenated layer bases
   const uint8_t* weight_data;  // matching per-base weights
   const uint32_t* layer_ends;  // per layer: end offset within window's span
+  // per layer: backbone span driving the SUBGRAPH-restricted alignment
+  // (begin << 16 | inclusive end; 0xFFFFFFFF = spans the window, full DP).
+  // CPU parity: layers not reaching within 1% of both window edges align
+  // against a subgraph of their backbone range (Window::generate_consensus);
+  // the device restricts DP rows to the [rank(begin), rank(end)] window —
+  // a capability the reference's cudapoa lacks (it aligns every layer to
+  // the full graph, which is what degrades its w=1000 GPU goldens).
+  const uint32_t* layer_spans;
   const uint32_t* layer_ends_index;  // per window: first index into layer_ends
   const PoaWindowDesc* windows;
 
