@@ -204,17 +204,24 @@ def test_gpu_stress_configs(racon, tmp_path_factory, fasta_reader):
     draft = list(fasta_reader(s1["layout"]).values())[0]
     assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth) * 0.2
 
-    # w=1000: windows at the 1023-column capacity edge (layers beyond 1023
-    # columns are skipped / fall back per reference semantics, so the
-    # residual is higher than at w=500 — the reference's own w=1000 golden
-    # is likewise its worst: 1289)
+    # w=1000: windows at the 1023-column capacity edge. Since the
+    # subgraph-restricted device alignment landed, the GPU tracks the CPU
+    # engine here (w=1000 is intrinsically the weakest config — the
+    # reference's own w=1000 golden is likewise its worst: CPU 1289, and
+    # its GPU one catastrophically so: 4168); assert closeness to the CPU
+    # result plus a sanity improvement over the draft.
     dw = tmp_path_factory.mktemp("wide")
     sw = synth.make_sample(dw, genome_bp=20000, coverage=25, seed=23)
     out = racon.polish(sw["reads"], sw["overlaps"], sw["layout"],
                        threads=4, poa_batches=1, window_length=1000)
+    cpu = racon.polish(sw["reads"], sw["overlaps"], sw["layout"],
+                       threads=4, window_length=1000)
     truth = list(fasta_reader(sw["reference"]).values())[0]
     draft = list(fasta_reader(sw["layout"]).values())[0]
-    assert racon.edit_distance(out[0][1], truth) < racon.edit_distance(draft, truth) * 0.6
+    ed_gpu = racon.edit_distance(out[0][1], truth)
+    ed_cpu = racon.edit_distance(cpu[0][1], truth)
+    assert ed_gpu < racon.edit_distance(draft, truth)
+    assert ed_gpu <= ed_cpu + max(20, ed_cpu // 5), (ed_cpu, ed_gpu)
 
     # high error: ~12% total error rate
     d2 = tmp_path_factory.mktemp("noisy")
