@@ -363,7 +363,7 @@ std::string AlignerBatch::cigar_of(uint32_t slot) const {
   return cigar;
 }
 
-uint32_t AlignerBatch::align_and_emit() {
+uint32_t AlignerBatch::align_and_emit(uint32_t window_length) {
   if (overlaps_.empty()) {
     return 0;
   }
@@ -385,7 +385,14 @@ uint32_t AlignerBatch::align_and_emit() {
         ++local_failed;  // empty CIGAR -> CPU pairwise fallback
         continue;
       }
-      overlaps_[i]->set_cigar(cigar);
+      overlaps_[i]->set_cigar(std::move(cigar));
+      if (window_length > 0) {
+        // walk into breaking points here (overlapped with other batches'
+        // GPU work); the polisher's final CPU pass then no-ops for this
+        // overlap and only realigns the skipped ones
+        overlaps_[i]->find_breaking_points_from_cigar(window_length);
+        overlaps_[i]->set_cigar(std::string());  // consumed; free the bytes
+      }
     }
     failed += local_failed;
   };

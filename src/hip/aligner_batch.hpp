@@ -52,9 +52,12 @@ class AlignerBatch {
 
   uint32_t size() const { return static_cast<uint32_t>(overlaps_.size()); }
 
-  // Runs the kernel (possibly several sub-launches) and writes CIGAR strings
-  // into the accepted overlaps; returns how many fell back (band-edge).
-  uint32_t align_and_emit();
+  // Runs the kernel (possibly several sub-launches), writes CIGAR strings
+  // into the accepted overlaps AND walks them into breaking points right on
+  // the emit threads (window_length > 0) — the walk overlaps the other
+  // batches' kernels instead of serializing in the post-GPU CPU pass.
+  // Returns how many overlaps fell back (band-edge -> CPU aligner).
+  uint32_t align_and_emit(uint32_t window_length = 0);
 
   void reset();
 

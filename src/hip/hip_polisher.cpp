@@ -227,7 +227,7 @@ class HipPolisher : public Polisher {
         if (pulled == 0) {
           return;
         }
-        skipped += batch->align_and_emit();
+        skipped += batch->align_and_emit(config_.window_length);
         t_gpu_ns += (clk::now() - t1).count();
       }
     };
