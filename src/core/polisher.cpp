@@ -364,6 +364,9 @@ void Polisher::find_overlap_breaking_points(std::vector<std::unique_ptr<Overlap>
   std::vector<std::future<void>> futures;
   futures.reserve(overlaps.size());
   for (uint64_t i = 0; i < overlaps.size(); ++i) {
+    if (!overlaps[i]->breaking_points().empty()) {
+      continue;  // already walked (GPU aligner emit path) — skip the no-op
+    }
     futures.emplace_back(thread_pool_->submit(
         [&](uint64_t j) { overlaps[j]->find_breaking_points(sequences_, config_.window_length); },
         i));

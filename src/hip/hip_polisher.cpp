@@ -323,30 +323,42 @@ class HipPolisher : public Polisher {
     auto worker = [&](hip::PoaBatch* batch) {
       using clk = std::chrono::steady_clock;
       std::vector<uint64_t> batch_indices;
+      // a window claimed from the queue but refused by a full batch is
+      // carried into the worker's next round (claim under the lock is just
+      // an index increment; the per-window bookkeeping — layer-order sort,
+      // measurements — runs outside so the workers do not serialize on it)
+      uint64_t carried = UINT64_MAX;
       while (true) {
         batch_indices.clear();
         batch->reset();
         auto t0 = clk::now();
-        {
-          std::lock_guard<std::mutex> lock(queue_mutex);
-          while (next_window < windows_.size()) {
-            auto& w = windows_[next_window];
-            if (w->num_layers() < 3) {
-              // backbone copy, never worth a GPU trip (reference semantics:
-              // consensus = backbone, unpolished)
-              w->set_consensus(std::string(w->sequence(0).first, w->sequence(0).second));
-              ++next_window;
-              continue;
+        while (true) {
+          uint64_t i;
+          if (carried != UINT64_MAX) {
+            i = carried;
+            carried = UINT64_MAX;
+          } else {
+            std::lock_guard<std::mutex> lock(queue_mutex);
+            if (next_window >= windows_.size()) {
+              break;
             }
-            bool never_fits = false;
-            if (batch->add_window(w, &never_fits)) {
-              batch_indices.emplace_back(next_window);
-              ++next_window;
-            } else if (never_fits) {
-              ++next_window;  // leave for the CPU fallback pass
-            } else {
-              break;  // batch full; leave remaining windows for the next round
-            }
+            i = next_window++;
+          }
+          auto& w = windows_[i];
+          if (w->num_layers() < 3) {
+            // backbone copy, never worth a GPU trip (reference semantics:
+            // consensus = backbone, unpolished)
+            w->set_consensus(std::string(w->sequence(0).first, w->sequence(0).second));
+            continue;
+          }
+          bool never_fits = false;
+          if (batch->add_window(w, &never_fits)) {
+            batch_indices.emplace_back(i);
+          } else if (never_fits) {
+            continue;  // leave for the CPU fallback pass
+          } else {
+            carried = i;  // batch full; keep the claim for the next round
+            break;
           }
         }
         auto t1 = clk::now();
