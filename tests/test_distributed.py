@@ -80,3 +80,38 @@ def test_bench_two_ranks_cpu(tmp_path):
     assert rec["value"] > 0
     # only rank 0 prints the JSON record
     assert not [l for l in outs[1].splitlines() if l.startswith("{")]
+
+
+COMM_BIG_WORKER = r'''
+import os, sys
+sys.path.insert(0, sys.argv[4])
+import _racon
+rank = int(sys.argv[1]); world = int(sys.argv[2]); port = int(sys.argv[3])
+_racon.comm_init(rank, world, "127.0.0.1", port, False)
+# multi-megabyte uneven payloads across several rounds (bench does one
+# gather per step; sizes vary as contigs change)
+for step in range(3):
+    payload = bytes([rank * 7 + step]) * (1_000_000 * (rank + 1) + step)
+    parts = _racon.comm_gather(payload, 0)
+    if rank == 0:
+        assert [len(p) for p in parts] == [1_000_000 * (r + 1) + step for r in range(world)]
+        assert all(p[0] == r * 7 + step for r, p in enumerate(parts))
+    _racon.comm_barrier()
+_racon.comm_finalize()
+print("COMM_BIG_OK")
+'''
+
+
+def test_comm_multi_round_large_payloads():
+    """World=4 TCP-plane gather with MB-scale uneven payloads over several
+    rounds (the shape of bench.py's per-step consensus gather)."""
+    procs = [
+        subprocess.Popen(
+            [sys.executable, "-c", COMM_BIG_WORKER, str(r), "4", "29537", str(REPO / "build")],
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+        for r in range(4)
+    ]
+    for p in procs:
+        out, err = p.communicate(timeout=180)
+        assert p.returncode == 0, err[-2000:]
+        assert "COMM_BIG_OK" in out

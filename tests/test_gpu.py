@@ -338,3 +338,30 @@ def test_gpu_real_scale_quality_and_determinism(racon, tmp_path_factory, fasta_r
                dict(threads=16, poa_batches=4, aligner_batches=4)):
         other = racon.polish(s["reads"], s["overlaps"], s["layout"], **kw)
         assert other == base, f"non-deterministic output under {kw}"
+
+
+def test_gpu_partial_layer_subgraph(racon):
+    """Partial layers (spans strictly inside the window) go through the
+    device subgraph restriction; results must track the CPU engine's
+    subgraph alignment closely (the memberships differ slightly — rank
+    window vs ancestor closure — so closeness, not bit-equality)."""
+    import random
+    rng = random.Random(9)
+    windows = []
+    for blen in (500, 800, 1000):
+        bb = "".join(rng.choice("ACGT") for _ in range(blen))
+        layers = [(bb, "!" * blen, 0, 0)]
+        for _ in range(12):  # full-span layers anchor the graph
+            layers.append((_mutate_rng(rng, bb, 0.02, 0.02, 0.02)[:1023], "", 0, blen))
+        for _ in range(12):  # partial layers: strictly interior spans
+            b = rng.randrange(20, blen // 2)
+            e = rng.randrange(blen // 2 + 10, blen - 10)
+            seg = _mutate_rng(rng, bb[b:e + 1], 0.02, 0.02, 0.02)[:1023]
+            layers.append((seg, "", b, e))
+        windows.append(layers)
+    cpu = racon.poa_windows_cpu(windows)
+    gpu = racon.poa_windows_gpu(windows)
+    for (c, _), (g, ok), layers in zip(cpu, gpu, windows):
+        assert ok
+        ed = racon.edit_distance(c, g)
+        assert ed <= max(4, len(c) // 100), (len(layers[0][0]), len(c), len(g), ed)
