@@ -179,43 +179,113 @@ void append_run(std::string& cigar, uint32_t n, char op) {
   cigar.append(buf, len);
 }
 
+constexpr int32_t kOutOfBand = (1 << 28);  // "unreached" marker in score columns
+
 // Scores of the FINAL column of NW(q[0..qn) vs t[0..tn)) for every query
-// prefix length 0..qn: out[i] = NW(q[0..i), t). out[0] = tn (boundary row).
-void nw_last_column(const char* q, uint32_t qn, const char* t, uint32_t tn,
+// prefix length 0..qn, computed inside a Ukkonen band of half-width k
+// around the main diagonal: out[i] = NW(q[0..i), t) for in-band rows,
+// kOutOfBand otherwise. out[0] = tn (the boundary row) when in band.
+//
+// Exactness: a cell with true value <= k satisfies |i - j| <= k, so it lies
+// in the band, and its optimal path does too; blocks entering the band get
+// the chain-from-above initialization (P = all ones continuing the vertical
+// +1 run), which upper-bounds the out-of-band cells — in-band values <= k
+// are therefore exact, which is all the Hirschberg crossing search reads
+// (crossing cells satisfy fwd + bwd == best, so both sides are <= best <= k).
+void nw_last_column(const char* q, uint32_t qn, const char* t, uint32_t tn, int64_t k,
                     std::vector<int32_t>* out) {
-  out->assign(qn + 1, 0);
-  (*out)[0] = static_cast<int32_t>(tn);
+  out->assign(qn + 1, kOutOfBand);
+  if (static_cast<int64_t>(tn) <= k + 0) {
+    (*out)[0] = static_cast<int32_t>(tn);
+  }
   if (qn == 0) {
     return;
   }
   Peq peq(q, qn, t, tn);
-  int nb = peq.num_blocks;
-  std::vector<Word> P(nb, ~Word(0));
-  std::vector<Word> M(nb, 0);
+  const int nb = peq.num_blocks;
+  std::vector<Word> P(nb), M(nb);
   std::vector<int32_t> bottom(nb);
-  for (int b = 0; b < nb; ++b) {
+
+  // row band for column j (1-based target prefix j): rows |i - j| <= k
+  auto lo_block = [&](int64_t j) {
+    const int64_t lo_row = j - k;  // first in-band row (1-based)
+    return static_cast<int>(std::max<int64_t>(0, (lo_row - 1) / kWordBits));
+  };
+  auto hi_block = [&](int64_t j) {
+    const int64_t hi_row = std::min<int64_t>(qn, j + k);
+    return static_cast<int>(std::min<int64_t>(nb - 1, (hi_row - 1) / kWordBits));
+  };
+
+  int first_b = 0;
+  int last_b = hi_block(1);
+  for (int b = first_b; b <= last_b; ++b) {
+    P[b] = ~Word(0);
+    M[b] = 0;
     bottom[b] = (b + 1) * kWordBits;
   }
   for (uint32_t c = 0; c < tn; ++c) {
+    const int64_t j = static_cast<int64_t>(c) + 1;
     const Word* eq_col =
         &peq.eq[static_cast<size_t>(peq.code_of[static_cast<unsigned char>(t[c])]) * nb];
+    // extend the bottom of the band with the virgin-chain initialization
+    const int want_hi = hi_block(j);
+    while (last_b < want_hi) {
+      ++last_b;
+      P[last_b] = ~Word(0);
+      M[last_b] = 0;
+      bottom[last_b] = bottom[last_b - 1] + kWordBits;
+    }
+    const int want_lo = lo_block(j);
+    // hin at the band top: out-of-band rows above follow the +1-per-column
+    // boundary chain, exactly like row 0
     int hin = 1;
-    for (int b = 0; b < nb; ++b) {
+    for (int b = std::max(first_b, want_lo); b <= last_b; ++b) {
       hin = myers_step(P[b], M[b], eq_col[b], hin);
       bottom[b] += hin;
     }
+    first_b = std::max(first_b, want_lo);
   }
-  // unpack the vertical delta bits into prefix scores (top-down accumulate)
-  int32_t v = static_cast<int32_t>(tn);
-  for (uint32_t i = 0; i < qn; ++i) {
-    const Word mask = Word(1) << (i % kWordBits);
-    const int b = static_cast<int>(i / kWordBits);
-    if (P[b] & mask) {
-      ++v;
-    } else if (M[b] & mask) {
-      --v;
+  // unpack in-band rows of the final column, walking up from block bottoms
+  const int flo = lo_block(tn), fhi = hi_block(tn);
+  const int64_t row_lo = std::max<int64_t>(0, static_cast<int64_t>(tn) - k);
+  const int64_t row_hi = std::min<int64_t>(qn, static_cast<int64_t>(tn) + k);
+  for (int b = std::max(first_b, flo); b <= std::min(last_b, fhi); ++b) {
+    int32_t v = bottom[b];
+    for (int bit = kWordBits - 1; bit >= 0; --bit) {
+      const int64_t row = static_cast<int64_t>(b) * kWordBits + bit + 1;  // 1-based
+      const Word mask = Word(1) << bit;
+      if (row <= static_cast<int64_t>(qn) && row >= row_lo && row <= row_hi) {
+        (*out)[row] = v;
+      }
+      if (P[b] & mask) {
+        --v;
+      } else if (M[b] & mask) {
+        ++v;
+      }
     }
-    (*out)[i + 1] = v;
+  }
+  if (row_lo == 0) {
+    (*out)[0] = static_cast<int32_t>(tn);
+  }
+}
+
+// Banded NW edit distance with iterative band doubling (edlib's top-level
+// strategy): exact, ~O((d/64) * tn) for distance d instead of O((qn/64) * tn).
+int64_t banded_distance(const char* q, uint32_t qn, const char* t, uint32_t tn) {
+  const int64_t dmin = qn > tn ? qn - tn : tn - qn;
+  int64_t k = std::max<int64_t>(kWordBits, dmin);
+  std::vector<int32_t> col;
+  while (true) {
+    nw_last_column(q, qn, t, tn, k, &col);
+    const int32_t v = col[qn];
+    if (v != kOutOfBand && v <= k) {
+      return v;
+    }
+    k *= 2;
+    if (k >= static_cast<int64_t>(qn) + tn) {
+      nw_last_column(q, qn, t, tn, k, &col);
+      return col[qn];
+    }
   }
 }
 
@@ -295,10 +365,13 @@ void obtain_ops(const char* q, const char* rq, uint32_t qn, const char* t, const
   const uint32_t right_w = tn - left_w;
 
   // middle-column scores from both directions: fwd[i] = NW(q[0..i), left
-  // half), bwd[j] = NW(q[qn-j..qn), right half)
+  // half), bwd[j] = NW(q[qn-j..qn), right half). Banded at k = the
+  // subproblem score: crossing cells satisfy fwd + bwd == score so both
+  // sides are <= score and in-band values at that magnitude are exact;
+  // out-of-band entries hold kOutOfBand and can never fake a crossing.
   std::vector<int32_t> fwd, bwd;
-  nw_last_column(q, qn, t, left_w, &fwd);
-  nw_last_column(rq, qn, rt, right_w, &bwd);
+  nw_last_column(q, qn, t, left_w, score, &fwd);
+  nw_last_column(rq, qn, rt, right_w, score, &bwd);
 
   // first (lowest) query row where the two halves meet at the optimum
   int64_t cross = -2;
@@ -331,7 +404,7 @@ int64_t edit_distance(const char* a, uint32_t a_len, const char* b, uint32_t b_l
   if (b_len == 0) {
     return a_len;
   }
-  return myers_nw(a, a_len, b, b_len, nullptr);
+  return banded_distance(a, a_len, b, b_len);
 }
 
 std::string align_global_cigar(const char* q, uint32_t q_len, const char* t, uint32_t t_len) {
@@ -345,7 +418,7 @@ std::string align_global_cigar(const char* q, uint32_t q_len, const char* t, uin
     return cigar;
   }
 
-  const int64_t score = myers_nw(q, q_len, t, t_len, nullptr);
+  const int64_t score = banded_distance(q, q_len, t, t_len);
 
   std::string rq(q, q + q_len), rt(t, t + t_len);
   std::reverse(rq.begin(), rq.end());
