@@ -12,7 +12,7 @@ trap 'rm -rf "$WORK"' EXIT
 python - "$WORK" <<'PY'
 import sys
 from racon_amd import synth
-synth.make_sample(sys.argv[1], genome_bp=200000, coverage=20, seed=5)
+synth.make_sample(sys.argv[1], genome_bp=1000000, coverage=20, seed=5)
 PY
 ./build/racon -t 4 -c 2 --cudaaligner-batches 2 "$WORK/reads.fasta" \
     "$WORK/overlaps.paf" "$WORK/layout.fasta" > "$WORK/run1.fasta"
