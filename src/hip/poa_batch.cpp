@@ -78,8 +78,15 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
 
   const size_t per_window = slab_bytes(limits_) + kSeqArenaPerWindow +
                             limits_.max_consensus * 3 + sizeof(PoaWindowDesc) + 1024;
+  // slab cap: 4096 default; RGA_POA_SLABS raises it for single-mega-batch
+  // experiments (all windows in one launch vs several contending launches)
+  static const size_t slab_cap = [] {
+    const char* e = getenv("RGA_POA_SLABS");
+    long v = e != nullptr ? atol(e) : 0;
+    return v > 0 ? static_cast<size_t>(v) : static_cast<size_t>(4096);
+  }();
   num_slabs_ = static_cast<uint32_t>(
-      std::min<size_t>(4096, std::max<size_t>(32, mem_budget / per_window)));
+      std::min<size_t>(slab_cap, std::max<size_t>(32, mem_budget / per_window)));
 
   // Exception safety: the destructor does not run when the constructor
   // throws, so free whatever was acquired before rethrowing (the polisher

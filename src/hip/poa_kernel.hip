@@ -898,7 +898,7 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
               bool p_has0 = true;
               if (banded) {
                 uint32_t pklo, pkhi;
-                band_chunks(p, slope16, c.bw, chunks, &pklo, &pkhi);
+                band_chunks(p - rlo, slope16, c.bw, chunks, &pklo, &pkhi);
                 plo = pklo * kLanes;  // valid cols: {0 if pklo==0} + [plo+1..phi]
                 phi = min(len, (pkhi + 1) * kLanes);
                 p_has0 = (pklo == 0);
