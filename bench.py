@@ -50,7 +50,7 @@ def main():
     ap.add_argument("--window", type=int, default=500)
     ap.add_argument("--threads", type=int, default=None,
                     help="CPU threads per rank (default: ncpu / (2*world))")
-    ap.add_argument("--poa-batches", type=int, default=8)
+    ap.add_argument("--poa-batches", type=int, default=4)
     ap.add_argument("--aligner-batches", type=int, default=4)
     ap.add_argument("--banded", action="store_true",
                     help="use -b static-band POA (approximation; off by default)")

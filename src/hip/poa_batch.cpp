@@ -83,7 +83,7 @@ PoaBatch::PoaBatch(int device, size_t mem_budget, int8_t match, int8_t mismatch,
   static const size_t slab_cap = [] {
     const char* e = getenv("RGA_POA_SLABS");
     long v = e != nullptr ? atol(e) : 0;
-    return v > 0 ? static_cast<size_t>(v) : static_cast<size_t>(4096);
+    return v > 0 ? static_cast<size_t>(v) : static_cast<size_t>(8192);
   }();
   num_slabs_ = static_cast<uint32_t>(
       std::min<size_t>(slab_cap, std::max<size_t>(32, mem_budget / per_window)));
