@@ -51,7 +51,7 @@ def main():
     ap.add_argument("--threads", type=int, default=None,
                     help="CPU threads per rank (default: ncpu / (2*world))")
     ap.add_argument("--poa-batches", type=int, default=4)
-    ap.add_argument("--aligner-batches", type=int, default=4)
+    ap.add_argument("--aligner-batches", type=int, default=6)
     ap.add_argument("--banded", action="store_true",
                     help="use -b static-band POA (approximation; off by default)")
     ap.add_argument("--cpu", action="store_true", help="force CPU path (debug)")
