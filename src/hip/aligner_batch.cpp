@@ -77,7 +77,10 @@ void AlignerBatch::allocate_arenas(size_t mem_budget) {
     tb_cap_u64_ = state / 20 * 16 / 8;  // 16/20 of state bytes as u64
     s_cap_i32_ = state / 20 * 4 / 4;    // 4/20 of state bytes as i32
 
-    max_waves = max_alignments_ / kLanes + 2;
+    // sized for the SMALLEST lanes-per-wave the launcher may pick (16 for
+    // small jobs), not kLanes — and the per-sub-launch descriptor slices
+    // accumulate across the whole run
+    max_waves = max_alignments_ / 16 + 2;
     total = 0;
     o_seqs = carve(seq_cap_);
     o_descs = carve(max_alignments_ * sizeof(AlnDesc));
@@ -250,6 +253,9 @@ void AlignerBatch::run() {
     lanes = 32;
   }
   uint32_t wave_begin = 0;  // in wave units
+  uint32_t wave_desc_off = 0;  // h_waves_/d_waves_ slice per sub-launch, so
+                               // sub-launches enqueue back-to-back with no
+                               // intervening stream sync
   const uint32_t num_waves_total = (na + lanes - 1) / lanes;
   std::vector<uint32_t> never_run;  // original indices of un-runnable waves
   auto wave_needs = [&](uint32_t w, uint64_t* peq_need, uint64_t* tb_need,
@@ -302,7 +308,7 @@ void AlignerBatch::run() {
       wd.s_off = s_off;
       wd.nb = nbm.first;
       wd.mmax = nbm.second;
-      h_waves_[launch_waves] = wd;
+      h_waves_[wave_desc_off + launch_waves] = wd;
       peq_off += peq_need;
       tb_off += tb_need;
       s_off += s_need;
@@ -310,15 +316,16 @@ void AlignerBatch::run() {
     }
     const uint32_t launch_align =
         std::min(na - wave_begin * lanes, launch_waves * lanes);
-    RGA_HIP_CHECK(hipMemcpyAsync(d_waves_, h_waves_, launch_waves * sizeof(AlnWaveDesc),
+    RGA_HIP_CHECK(hipMemcpyAsync(d_waves_ + wave_desc_off, h_waves_ + wave_desc_off,
+                                 launch_waves * sizeof(AlnWaveDesc),
                                  hipMemcpyHostToDevice, s));
     AlnDeviceArena launch_arena = arena_;
     launch_arena.order = d_order_ + wave_begin * lanes;
-    launch_arena.waves = d_waves_;
+    launch_arena.waves = d_waves_ + wave_desc_off;
     launch_arena.lanes_per_wave = lanes;
     launch_aligner_kernel(launch_arena, launch_waves, launch_align, K, stream_);
-    RGA_HIP_CHECK(hipStreamSynchronize(s));
     wave_begin += launch_waves;
+    wave_desc_off += launch_waves;
   }
 
   RGA_HIP_CHECK(hipMemcpyAsync(h_path_, arena_.path, path_bytes_, hipMemcpyDeviceToHost, s));
