@@ -252,6 +252,16 @@ void AlignerBatch::run() {
   } else if (na < 49152) {
     lanes = 32;
   }
+  // tuning override (RGA_ALN_LANES in {16,32,64}): alignments packed per
+  // 64-lane wave — fewer per wave = more interleavable waves per CU at the
+  // cost of idle lanes
+  static const long lanes_env = [] {
+    const char* e = getenv("RGA_ALN_LANES");
+    return e != nullptr ? atol(e) : 0;
+  }();
+  if (lanes_env == 16 || lanes_env == 32 || lanes_env == 64) {
+    lanes = static_cast<uint32_t>(lanes_env);
+  }
   uint32_t wave_begin = 0;  // in wave units
   uint32_t wave_desc_off = 0;  // h_waves_/d_waves_ slice per sub-launch, so
                                // sub-launches enqueue back-to-back with no
