@@ -355,31 +355,40 @@ __device__ void add_alignment_d(WindowCtx& c, const uint8_t* seq,
 // Kahn topological sort (FIFO, deterministic) over the LDS mirrors. The
 // FIFO queue IS the final topological order (s.u.kahn.queue); rank goes to
 // the global slab (read lane-parallel by build_row_desc).
+// Cooperative: the in-degree staging and the rank writeback are
+// lane-parallel; only the FIFO core (whose order is the deterministic
+// topological order) stays on lane 0. Call from ALL lanes.
 template <class SH>
-__device__ void topo_sort_d(WindowCtx& c, SH& s) {
-  uint32_t n = c.num_nodes;
-  for (uint32_t i = 0; i < n; ++i) {
+__device__ void topo_sort_d(WindowCtx& c, SH& s, int lane) {
+  const uint32_t n = c.num_nodes;
+  for (uint32_t i = lane; i < n; i += kLanes) {
     s.u.kahn.work[i] = c.in_cnt[i];
   }
-  uint32_t qhead = 0, qtail = 0;
-  for (uint32_t i = 0; i < n; ++i) {
-    if (s.u.kahn.work[i] == 0) {
-      s.u.kahn.queue[qtail++] = static_cast<uint16_t>(i);
+  __syncthreads();
+  if (lane == 0) {
+    uint32_t qhead = 0, qtail = 0;
+    for (uint32_t i = 0; i < n; ++i) {
+      if (s.u.kahn.work[i] == 0) {
+        s.u.kahn.queue[qtail++] = static_cast<uint16_t>(i);
+      }
     }
-  }
-  while (qhead < qtail) {
-    uint16_t u = s.u.kahn.queue[qhead++];
-    uint32_t nout = c.out_cnt[u];
-    for (uint32_t e = 0; e < nout; ++e) {
-      uint16_t v = out_edge_of(c, u, e);
-      if (--s.u.kahn.work[v] == 0) {
-        s.u.kahn.queue[qtail++] = v;
+    while (qhead < qtail) {
+      uint16_t u = s.u.kahn.queue[qhead++];
+      uint32_t nout = c.out_cnt[u];
+      for (uint32_t e = 0; e < nout; ++e) {
+        uint16_t v = out_edge_of(c, u, e);
+        if (--s.u.kahn.work[v] == 0) {
+          s.u.kahn.queue[qtail++] = v;
+        }
       }
     }
   }
-  for (uint32_t r = 0; r < qtail; ++r) {
+  __syncthreads();
+  for (uint32_t r = lane; r < n; r += kLanes) {
     c.rank[s.u.kahn.queue[r]] = static_cast<uint16_t>(r);
   }
+  __syncthreads();  // rank stores -> visible to the cross-lane readers
+                    // (build_row_desc, the DP's pred-rank loads)
 }
 
 // Heaviest-bundle consensus (mirrors Graph::traverse_heaviest_bundle).
@@ -1206,10 +1215,6 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
         add_alignment_d(c, seq, wts, len, aln_len);
       }
       t_add += lap();
-      if (c.status == kPoaOk) {
-        topo_sort_d(c, s);
-      }
-      t_topo += lap();
     }
 
     // lane 0's graph updates must be visible to the whole wave
@@ -1218,6 +1223,14 @@ __global__ void poa_window_kernel(PoaDeviceArena a, uint32_t window_base,
     c.num_nodes = __shfl(c.num_nodes, 0, kLanes);
     c.seqs_in_graph = __shfl(c.seqs_in_graph, 0, kLanes);
     c.status = __shfl(c.status, 0, kLanes);
+
+    (void)lap();
+    if (c.status == kPoaOk) {
+      topo_sort_d(c, s, lane);  // cooperative (internal barriers)
+    }
+    if (lane == 0) {
+      t_topo += lap();
+    }
 
     // rebuild the packed row descriptors for the grown graph, lane-parallel
     (void)lap();
