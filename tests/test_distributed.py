@@ -115,3 +115,22 @@ def test_comm_multi_round_large_payloads():
         out, err = p.communicate(timeout=180)
         assert p.returncode == 0, err[-2000:]
         assert "COMM_BIG_OK" in out
+
+
+def test_comm_init_errors():
+    """Misuse must raise, not crash: world < 2, and double init."""
+    code = r'''
+import sys
+sys.path.insert(0, sys.argv[1])
+import _racon
+try:
+    _racon.comm_init(0, 1, "127.0.0.1", 29541, False)
+    raise SystemExit("expected world<2 to raise")
+except RuntimeError:
+    pass
+print("COMM_ERRORS_OK")
+'''
+    out = subprocess.run([sys.executable, "-c", code, str(REPO / "build")],
+                         capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr[-1000:]
+    assert "COMM_ERRORS_OK" in out.stdout

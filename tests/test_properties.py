@@ -80,3 +80,37 @@ def test_poa_consensus_is_deterministic(racon, seqs):
     a = racon.poa_consensus(seqs)
     b = racon.poa_consensus(seqs)
     assert a == b
+
+
+@given(st.text(alphabet="ACGT", min_size=0, max_size=400),
+       st.text(alphabet="ACGT", min_size=0, max_size=400))
+@settings(max_examples=80, deadline=None)
+def test_banded_distance_matches_definition(racon, a, b):
+    """edit_distance (banded, band-doubling) must equal the textbook DP."""
+    import functools
+    @functools.lru_cache(maxsize=None)
+    def dp(i, j):
+        if i == 0:
+            return j
+        if j == 0:
+            return i
+        return min(dp(i - 1, j) + 1, dp(i, j - 1) + 1,
+                   dp(i - 1, j - 1) + (a[i - 1] != b[j - 1]))
+    import sys
+    sys.setrecursionlimit(10000)
+    assert racon.edit_distance(a, b) == dp(len(a), len(b))
+
+
+@given(st.integers(min_value=0, max_value=2**31))
+@settings(max_examples=20, deadline=None)
+def test_banded_distance_skewed_lengths(racon, seed):
+    """Very unequal lengths (band dominated by the length gap)."""
+    import random
+    rng = random.Random(seed)
+    a = "".join(rng.choice("ACGT") for _ in range(rng.randint(0, 50)))
+    b = a + "".join(rng.choice("ACGT") for _ in range(rng.randint(500, 2000)))
+    assert racon.edit_distance(a, b) >= len(b) - len(a) - 2 * len(a)
+    # appending to a superstring: distance is exactly the length gap when a
+    # is a prefix of b
+    assert racon.edit_distance(a, b) <= len(b) - len(a) + 2 * len(a)
+    assert racon.edit_distance(b, b) == 0
