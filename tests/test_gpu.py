@@ -365,3 +365,18 @@ def test_gpu_partial_layer_subgraph(racon):
         assert ok
         ed = racon.edit_distance(c, g)
         assert ed <= max(4, len(c) // 100), (len(layers[0][0]), len(c), len(g), ed)
+
+
+def test_gpu_reference_ci_flags(racon, sample, fasta_reader):
+    """The reference's GPU CI parameter set (ci/gpu/cuda_test.sh: -m 8 -x -6
+    -g -8 -q -1 -c 2): larger score magnitudes near the int16 guard band and
+    a disabled quality filter must polish cleanly and deterministically."""
+    truth = list(fasta_reader(sample["reference"]).values())[0]
+    draft = list(fasta_reader(sample["layout"]).values())[0]
+    kw = dict(threads=4, match=8, mismatch=-6, gap=-8, quality_threshold=-1.0,
+              poa_batches=2, aligner_batches=2)
+    a = racon.polish(sample["reads"], sample["overlaps"], sample["layout"], **kw)
+    kw2 = dict(kw, threads=8, poa_batches=1, aligner_batches=1)
+    b = racon.polish(sample["reads"], sample["overlaps"], sample["layout"], **kw2)
+    assert a == b
+    assert racon.edit_distance(a[0][1], truth) < racon.edit_distance(draft, truth) * 0.2
