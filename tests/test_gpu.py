@@ -96,11 +96,13 @@ def test_gpu_matches_cpu_closely(racon, sample):
     gpu = racon.polish(sample["reads"], sample["overlaps"], sample["layout"],
                        threads=4, poa_batches=1)
     assert len(cpu) == len(gpu)
-    # GPU tie-breaking (Kahn topological order) may differ from the CPU DFS
-    # order; the reference pins different goldens for each path too. Demand
-    # near-identity: < 0.5% divergence.
+    # With the device subgraph restriction the engines agree bit-for-bit on
+    # this sample (measured ed == 0 end to end); GPU tie-breaking (Kahn
+    # order vs CPU DFS) may still differ on other inputs, so allow a few
+    # bases of slack rather than exact equality. (The reference pins GPU
+    # goldens ~5% away from its CPU ones — racon_test.cpp:312.)
     ed = racon.edit_distance(cpu[0][1], gpu[0][1])
-    assert ed < 0.005 * len(cpu[0][1]), ed
+    assert ed <= 8, ed
 
 
 def test_gpu_banded_poa(racon, sample, fasta_reader):
