@@ -78,6 +78,12 @@ def main():
                         help="banded approximation for GPU POA alignment")
     parser.add_argument("--cudaaligner-band-width", default=0,
                         help="band width for GPU alignment (0 = auto)")
+    parser.add_argument("--in-process", action="store_true",
+                        help="polish every --split chunk inside ONE process via the "
+                             "_racon module instead of one racon subprocess per chunk: "
+                             "the GPU batch arenas are built once and reused across "
+                             "chunks (~2x on genome-scale GPU splits); default stays "
+                             "one-subprocess-per-chunk for reference parity")
     args = parser.parse_args()
 
     racon = find_binary("RACON_BIN", "racon")
@@ -117,6 +123,31 @@ def main():
                 sys.exit(1)
         else:
             chunks = [targets]
+
+        if args.in_process:
+            sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                            "..", "build"))
+            import _racon
+            for chunk in chunks:
+                eprint(f"[racon_wrapper] polishing {os.path.basename(chunk)} (in-process)")
+                out = _racon.polish(
+                    sequences, overlaps, chunk,
+                    fragment_correction=args.fragment_correction,
+                    window_length=int(args.window_length),
+                    quality_threshold=float(args.quality_threshold),
+                    error_threshold=float(args.error_threshold),
+                    trim=not args.no_trimming,
+                    match=int(args.match), mismatch=int(args.mismatch),
+                    gap=int(args.gap), threads=int(args.threads),
+                    poa_batches=int(args.cudapoa_batches),
+                    banded_poa=args.cuda_banded_alignment,
+                    aligner_batches=int(args.cudaaligner_batches),
+                    aligner_band_width=int(args.cudaaligner_band_width),
+                    include_unpolished=args.include_unpolished)
+                for name, seq in out:
+                    sys.stdout.write(f">{name}\n{seq}\n")
+                sys.stdout.flush()
+            return
 
         cmd = [racon]
         if args.include_unpolished:

@@ -137,3 +137,20 @@ def test_rampler_edge_cases(tmp_path, rampler):
     assert r.returncode == 0
     assert (tmp_path / "single_0.fasta").exists()
     assert not (tmp_path / "single_1.fasta").exists()
+
+
+def test_wrapper_in_process_split_matches_subprocess(racon, sample, tmp_path):
+    """--in-process chunk polishing (single process, pooled GPU arenas) must
+    be byte-identical to the reference-parity one-subprocess-per-chunk mode."""
+    import subprocess
+    import sys as _sys
+    from pathlib import Path
+    repo = Path(__file__).resolve().parent.parent
+    base = [_sys.executable, str(repo / "scripts" / "racon_wrapper.py"),
+            "--split", "8000", "-t", "4",
+            sample["reads"], sample["overlaps"], sample["layout"]]
+    a = subprocess.run(base, capture_output=True, text=True, timeout=600)
+    assert a.returncode == 0, a.stderr[-800:]
+    b = subprocess.run(base + ["--in-process"], capture_output=True, text=True, timeout=600)
+    assert b.returncode == 0, b.stderr[-800:]
+    assert a.stdout == b.stdout
