@@ -341,6 +341,13 @@ def test_gpu_real_scale_quality_and_determinism(racon, tmp_path_factory, fasta_r
         other = racon.polish(s["reads"], s["overlaps"], s["layout"], **kw)
         assert other == base, f"non-deterministic output under {kw}"
 
+    # engine faithfulness at scale: the GPU path tracks the CPU engine
+    # (subgraph semantics on device; GPU CIGARs may differ from the CPU
+    # aligner's equal-cost choices, so near- rather than bit-equality)
+    cpu = racon.polish(s["reads"], s["overlaps"], s["layout"], threads=16)
+    ed_cpu_gpu = racon.edit_distance(cpu[0][1], base[0][1])
+    assert ed_cpu_gpu < 0.0005 * len(cpu[0][1]), ed_cpu_gpu
+
 
 def test_gpu_partial_layer_subgraph(racon):
     """Partial layers (spans strictly inside the window) go through the
